@@ -1,0 +1,143 @@
+"""Prototxt parsing: the reference's actual usage files must round-trip."""
+
+import math
+import os
+
+from npairloss_amd.config.params import (
+    MiningMethod,
+    MiningRegion,
+    NPairLossConfig,
+    SolverConfig,
+    parse_net_prototxt,
+)
+from npairloss_amd.config.prototxt import parse_prototxt, format_prototxt
+
+SOLVER_TEXT = """
+net: "./conf_same_veri/def.prototxt"
+test_iter: 2000
+test_interval: 2000 # 10000
+test_initialization:true # false
+display: 100
+average_loss: 100
+base_lr: 0.001
+lr_policy: "step"
+stepsize: 10000
+gamma: 0.5
+#power: 1
+max_iter: 2000000
+momentum: 0.9
+weight_decay: 0.00002
+snapshot: 5000
+snapshot_prefix: "./snap/googlenet_"
+solver_mode: GPU
+"""
+
+LOSS_LAYER_TEXT = """
+layer {
+    bottom: "loss3/pool5/7x7_s1/norm"
+    bottom: "label_type_mb"
+    name: "loss3/type_mb"
+    type: "NPairMultiClassLoss"
+    top: "loss3/type_npair_mc"
+    top: "loss3/type_npair_mc_retrieve_top1"
+    top: "loss3/type_npair_mc_retrieve_top5"
+    top: "loss3/type_npair_mc_retrieve_top10"
+    top: "loss3/feature_asum"
+    loss_weight: 1
+    loss_weight: 1
+    loss_weight: 1
+    loss_weight: 1
+    loss_weight: 1
+    npair_loss_param {
+        margin_ident: 0.0
+        margin_diff: -0.05
+        identsn: -0.0
+        diffsn: -0.3 # ignored for absolute selection
+        ap_mining_region: GLOBAL
+        ap_mining_method: RELATIVE_HARD
+        an_mining_region: LOCAL
+        an_mining_method: HARD
+    }
+}
+"""
+
+
+def test_solver_parse():
+    s = SolverConfig.from_prototxt(SOLVER_TEXT)
+    assert s.base_lr == 0.001
+    assert s.lr_policy == "step"
+    assert s.stepsize == 10000
+    assert s.gamma == 0.5
+    assert s.momentum == 0.9
+    assert s.weight_decay == 2e-5
+    assert s.max_iter == 2000000
+    assert s.snapshot == 5000
+    assert s.snapshot_prefix == "./snap/googlenet_"
+    assert s.test_interval == 2000 and s.test_iter == 2000
+    assert s.display == 100 and s.average_loss == 100
+    # step LR policy: x0.5 every 10k
+    assert s.lr_at(0) == 0.001
+    assert s.lr_at(9999) == 0.001
+    assert s.lr_at(10000) == 0.0005
+    assert s.lr_at(25000) == 0.001 * 0.5 ** 2
+
+
+def test_loss_layer_parse():
+    net = parse_net_prototxt(LOSS_LAYER_TEXT)
+    assert len(net.layers) == 1
+    layer = net.layers[0]
+    assert layer.type == "NPairMultiClassLoss"
+    assert layer.bottoms == ["loss3/pool5/7x7_s1/norm", "label_type_mb"]
+    assert len(layer.tops) == 5
+    assert layer.loss_weights == [1.0] * 5
+    cfg = NPairLossConfig.from_message(layer.raw.get("npair_loss_param"))
+    assert cfg.margin_ident == 0.0
+    assert cfg.margin_diff == float(__import__("numpy").float32(-0.05))
+    assert cfg.ap_mining_region == MiningRegion.GLOBAL
+    assert cfg.ap_mining_method == MiningMethod.RELATIVE_HARD
+    assert cfg.an_mining_region == MiningRegion.LOCAL
+    assert cfg.an_mining_method == MiningMethod.HARD
+    # -0.0 must keep its sign bit irrelevant: >= 0 branch
+    assert cfg.identsn == 0.0 and cfg.identsn >= 0
+
+
+def test_defaults_match_proto():
+    cfg = NPairLossConfig()
+    assert cfg.margin_ident == 0 and cfg.margin_diff == 0
+    assert cfg.identsn == -1.0 and cfg.diffsn == -1.0
+    assert cfg.ap_mining_region == MiningRegion.LOCAL
+    assert cfg.ap_mining_method == MiningMethod.RAND
+    assert cfg.an_mining_region == MiningRegion.LOCAL
+    assert cfg.an_mining_method == MiningMethod.RAND
+
+
+def test_roundtrip():
+    msg = parse_prototxt(LOSS_LAYER_TEXT)
+    text = format_prototxt(msg)
+    msg2 = parse_prototxt(text)
+    l1 = msg.get("layer")
+    l2 = msg2.get("layer")
+    assert l1.get_all("loss_weight") == l2.get_all("loss_weight")
+    assert l1.get("npair_loss_param").get("ap_mining_region") == "GLOBAL"
+    assert l2.get("npair_loss_param").get("ap_mining_region") == "GLOBAL"
+
+
+def test_reference_def_prototxt_parses_if_present():
+    path = "/root/reference/usage/def.prototxt"
+    if not os.path.exists(path):
+        return
+    # The file contains a literal "." elided-section marker on three lines;
+    # strip those (they are not valid prototxt).
+    text = "\n".join(l for l in open(path).read().splitlines() if l.strip() != ".")
+    net = parse_net_prototxt(text)
+    types = [l.type for l in net.layers]
+    assert "NPairMultiClassLoss" in types
+    assert "MultibatchData" in types
+    loss = net.find("NPairMultiClassLoss")[0]
+    cfg = NPairLossConfig.from_message(loss.raw.get("npair_loss_param"))
+    assert cfg.ap_mining_method == MiningMethod.RELATIVE_HARD
+    mb = net.find("MultibatchData")[0]
+    p = mb.raw.get("multi_batch_data_param")
+    assert p.get("batch_size") == 120
+    assert p.get("identity_num_per_batch") == 60
+    assert p.get("img_num_per_identity") == 2
