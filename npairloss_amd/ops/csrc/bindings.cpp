@@ -1,0 +1,57 @@
+// Python bindings for the npairloss_amd gfx950 HIP kernels.
+
+#include <torch/extension.h>
+
+#include <vector>
+
+// npair_rows.hip
+std::vector<torch::Tensor> rowstats(torch::Tensor S, torch::Tensor lab_l,
+                                    torch::Tensor lab_g, int64_t rank);
+std::vector<torch::Tensor> fused_fwd(torch::Tensor S, torch::Tensor lab_l,
+                                     torch::Tensor lab_g, int64_t rank,
+                                     torch::Tensor thr_p, torch::Tensor thr_n,
+                                     torch::Tensor max_all, double margin_ident,
+                                     double margin_diff, int64_t ap_method,
+                                     int64_t an_method);
+torch::Tensor bwd_weights(torch::Tensor S, torch::Tensor lab_l,
+                          torch::Tensor lab_g, int64_t rank,
+                          torch::Tensor thr_p, torch::Tensor thr_n,
+                          torch::Tensor max_all, torch::Tensor loss_ident,
+                          torch::Tensor loss_sum, double margin_ident,
+                          double margin_diff, int64_t ap_method,
+                          int64_t an_method, double scale);
+torch::Tensor recall_hits(torch::Tensor S, torch::Tensor lab_l,
+                          torch::Tensor lab_g, int64_t rank,
+                          std::vector<int64_t> ks);
+
+// sort_select.hip
+torch::Tensor local_relative_thr(torch::Tensor S, torch::Tensor lab_l,
+                                 torch::Tensor lab_g, int64_t rank,
+                                 bool use_same, double sn);
+torch::Tensor global_relative_thr(torch::Tensor S, torch::Tensor lab_l,
+                                  torch::Tensor lab_g, int64_t rank,
+                                  bool use_same, double sn);
+
+// l2norm.hip
+std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x);
+torch::Tensor l2norm_bwd(torch::Tensor y, torch::Tensor inv_norm, torch::Tensor dy);
+
+// gemm_f32.hip
+torch::Tensor sim_gemm_nt(torch::Tensor F_l, torch::Tensor F_g);
+torch::Tensor gemm_nn(torch::Tensor A, torch::Tensor B);
+torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "npairloss_amd gfx950 HIP kernels";
+  m.def("rowstats", &rowstats, "per-query min_within/max_between/max_all");
+  m.def("fused_fwd", &fused_fwd, "fused mask+select+count+LSE loss pass");
+  m.def("bwd_weights", &bwd_weights, "fused (-p1+p2+p3) backward weights");
+  m.def("recall_hits", &recall_hits, "Recall@k hit counts");
+  m.def("local_relative_thr", &local_relative_thr, "per-row bitonic order-statistic threshold");
+  m.def("global_relative_thr", &global_relative_thr, "device radix-select order-statistic threshold");
+  m.def("l2norm_fwd", &l2norm_fwd, "row L2 normalize forward");
+  m.def("l2norm_bwd", &l2norm_bwd, "row L2 normalize backward");
+  m.def("sim_gemm_nt", &sim_gemm_nt, "fp32 MFMA similarity GEMM (A @ B^T)");
+  m.def("gemm_nn", &gemm_nn, "fp32 MFMA GEMM A @ B");
+  m.def("gemm_tn", &gemm_tn, "fp32 MFMA GEMM A^T @ B");
+}
