@@ -1,0 +1,132 @@
+"""GoogLeNet (Inception v1) embedding backbone, Caffe-layout.
+
+The reference's usage/def.prototxt trains a GoogLeNet whose conv stack is
+elided in the file; the named endpoints (conv1/7x7_s2 at def.prototxt:86,
+pool5/7x7_s1 at :115-117) identify it as the standard BVLC GoogLeNet v1.
+This is a from-scratch PyTorch implementation in the Caffe layout
+(LRN after pool1/conv2, ceil-mode pooling, ReLU everywhere, 7x7 average
+pool -> 1024-d embedding) with the original Caffe layer names preserved in
+`caffe_names()` so `.caffemodel` checkpoints load by name
+(utils/checkpoint.py).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Tuple
+
+import torch
+from torch import nn
+
+
+class ConvReLU(nn.Module):
+    def __init__(self, cin: int, cout: int, k: int, stride: int = 1, pad: int = 0):
+        super().__init__()
+        self.conv = nn.Conv2d(cin, cout, k, stride=stride, padding=pad, bias=True)
+        self.relu = nn.ReLU(inplace=True)
+
+    def forward(self, x):
+        return self.relu(self.conv(x))
+
+
+class Inception(nn.Module):
+    """The 4-branch inception block: 1x1 | 1x1->3x3 | 1x1->5x5 | pool->1x1."""
+
+    def __init__(self, cin: int, c1: int, c3r: int, c3: int, c5r: int, c5: int, cp: int):
+        super().__init__()
+        self.b1 = ConvReLU(cin, c1, 1)
+        self.b3_reduce = ConvReLU(cin, c3r, 1)
+        self.b3 = ConvReLU(c3r, c3, 3, pad=1)
+        self.b5_reduce = ConvReLU(cin, c5r, 1)
+        self.b5 = ConvReLU(c5r, c5, 5, pad=2)
+        self.pool = nn.MaxPool2d(3, stride=1, padding=1, ceil_mode=True)
+        self.pool_proj = ConvReLU(cin, cp, 1)
+
+    def forward(self, x):
+        return torch.cat(
+            [self.b1(x), self.b3(self.b3_reduce(x)), self.b5(self.b5_reduce(x)),
+             self.pool_proj(self.pool(x))], dim=1)
+
+
+# (c1, c3r, c3, c5r, c5, cp) per block — standard BVLC GoogLeNet v1
+_INCEPTION_CFG = {
+    "3a": (64, 96, 128, 16, 32, 32),
+    "3b": (128, 128, 192, 32, 96, 64),
+    "4a": (192, 96, 208, 16, 48, 64),
+    "4b": (160, 112, 224, 24, 64, 64),
+    "4c": (128, 128, 256, 24, 64, 64),
+    "4d": (112, 144, 288, 32, 64, 64),
+    "4e": (256, 160, 320, 32, 128, 128),
+    "5a": (256, 160, 320, 32, 128, 128),
+    "5b": (384, 192, 384, 48, 128, 128),
+}
+_INCEPTION_IN = {"3a": 192, "3b": 256, "4a": 480, "4b": 512, "4c": 512,
+                 "4d": 512, "4e": 528, "5a": 832, "5b": 832}
+
+
+class GoogLeNet(nn.Module):
+    """Input B x 3 x 224 x 224 -> 1024-d embedding (pool5/7x7_s1 output,
+    the reference's embedding endpoint, def.prototxt:115-120)."""
+
+    embed_dim = 1024
+
+    def __init__(self, dropout: float = 0.4):
+        super().__init__()
+        self.conv1 = ConvReLU(3, 64, 7, stride=2, pad=3)
+        self.pool1 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
+        self.norm1 = nn.LocalResponseNorm(5, alpha=1e-4, beta=0.75)
+        self.conv2_reduce = ConvReLU(64, 64, 1)
+        self.conv2 = ConvReLU(64, 192, 3, pad=1)
+        self.norm2 = nn.LocalResponseNorm(5, alpha=1e-4, beta=0.75)
+        self.pool2 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
+        self.inception = nn.ModuleDict({
+            name: Inception(_INCEPTION_IN[name], *cfg) for name, cfg in _INCEPTION_CFG.items()
+        })
+        self.pool3 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
+        self.pool4 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
+        self.pool5 = nn.AdaptiveAvgPool2d(1)  # 7x7 avg at 224 input
+        self.dropout = nn.Dropout(dropout)
+        self._init_weights()
+
+    def _init_weights(self):
+        for m in self.modules():
+            if isinstance(m, nn.Conv2d):
+                nn.init.xavier_uniform_(m.weight)
+                if m.bias is not None:
+                    nn.init.constant_(m.bias, 0.2)  # def.prototxt:109-112 filler
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.norm1(self.pool1(self.conv1(x)))
+        x = self.pool2(self.norm2(self.conv2(self.conv2_reduce(x))))
+        x = self.inception["3a"](x)
+        x = self.inception["3b"](x)
+        x = self.pool3(x)
+        for name in ("4a", "4b", "4c", "4d", "4e"):
+            x = self.inception[name](x)
+        x = self.pool4(x)
+        x = self.inception["5a"](x)
+        x = self.inception["5b"](x)
+        x = self.pool5(x)
+        x = torch.flatten(x, 1)
+        return self.dropout(x)
+
+    # -- Caffe name mapping (for .caffemodel loading) ------------------------
+
+    def caffe_names(self) -> Dict[str, nn.Conv2d]:
+        """Caffe layer name -> conv module holding its (weight, bias)."""
+        m: Dict[str, nn.Conv2d] = {
+            "conv1/7x7_s2": self.conv1.conv,
+            "conv2/3x3_reduce": self.conv2_reduce.conv,
+            "conv2/3x3": self.conv2.conv,
+        }
+        branch_names = {
+            "b1": "1x1",
+            "b3_reduce": "3x3_reduce",
+            "b3": "3x3",
+            "b5_reduce": "5x5_reduce",
+            "b5": "5x5",
+            "pool_proj": "pool_proj",
+        }
+        for blk, mod in self.inception.items():
+            for attr, suffix in branch_names.items():
+                m[f"inception_{blk}/{suffix}"] = getattr(mod, attr).conv
+        return m

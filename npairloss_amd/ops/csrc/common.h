@@ -115,16 +115,18 @@ DEVINL bool select_neg(float s, float thr, int method) {
   }
 }
 
-// Relative-order-statistic index (reference .cu:285-287), computed in
-// double from the float32 sn; clamped to [0, size-1] (reference is UB out
-// of range).  Returns -1 for an empty list.
+// Relative-order-statistic index (reference .cu:285-287).  The reference's
+// C expression `size_t - 1 + float_sn * size_t` promotes to FLOAT32, so we
+// must compute in float32 too (double diverges when sn*size is near an
+// integer).  Clamped to [0, size-1] (reference is UB out of range).
+// Returns -1 for an empty list.
 DEVINL long long relative_index(float sn, long long size) {
   if (size <= 0) return -1;
   long long pos;
   if (sn >= 0.0f) {
     pos = size - 1 - (long long)sn;
   } else {
-    pos = (long long)((double)size - 1.0 + (double)sn * (double)size);
+    pos = (long long)((float)(size - 1) + sn * (float)size);
   }
   if (pos < 0) pos = 0;
   if (pos > size - 1) pos = size - 1;

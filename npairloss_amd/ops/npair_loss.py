@@ -66,16 +66,16 @@ def _row_stats(S: torch.Tensor, same: torch.Tensor, diff: torch.Tensor):
 
 
 def _rel_index(sn: float, sizes: torch.Tensor) -> torch.Tensor:
-    """Reference .cu:285-287 arithmetic, vectorized; clamped to [0, size-1]."""
-    sizes_d = sizes.to(torch.float64)
-    if sn >= 0 or (sn == 0 and math.copysign(1, sn) > 0):  # -0.0 handled below
-        pos = sizes_d - 1 - float(int(sn))
+    """Reference .cu:285-287 arithmetic, vectorized; clamped to [0, size-1].
+    The reference's `size_t - 1 + float * size_t` promotes to FLOAT32, so
+    the negative-sn branch is computed in float32 (see oracle.relative_index).
+    -0.0 >= 0 is True in both C and Python, matching the reference branch."""
+    if sn >= 0:
+        pos = (sizes - 1 - int(sn)).to(torch.float32)
     else:
-        pos = torch.trunc(sizes_d - 1 + sn * sizes_d)
-    # C semantics: sn >= 0 uses the "absolute count from top" branch and
-    # -0.0 >= 0 is True there; Python float -0.0 >= 0 is also True, so the
-    # branch above already matches.
-    return pos.clamp(min=0).minimum(sizes_d - 1).to(torch.long)
+        sizes_f = sizes.to(torch.float32)
+        pos = torch.trunc((sizes_f - 1) + float(sn) * sizes_f)
+    return pos.clamp(min=0).minimum((sizes - 1).clamp(min=0).to(torch.float32)).to(torch.long)
 
 
 def _local_relative_thr(S: torch.Tensor, mask: torch.Tensor, sn: float) -> torch.Tensor:
@@ -89,16 +89,14 @@ def _local_relative_thr(S: torch.Tensor, mask: torch.Tensor, sn: float) -> torch
 
 
 def _global_relative_thr(S: torch.Tensor, mask: torch.Tensor, sn: float) -> torch.Tensor:
+    from .oracle import relative_index
+
     vals = S[mask]
     n = vals.numel()
     if n == 0:
         return S.new_full((), NEG_INF)
     sorted_vals, _ = vals.sort()
-    if sn >= 0:
-        pos = n - 1 - int(sn)
-    else:
-        pos = int(n - 1 + sn * n)
-    pos = min(max(pos, 0), n - 1)
+    pos = relative_index(sn, n)
     thr = sorted_vals[pos]
     return torch.where(thr >= 0, thr, torch.full_like(thr, NEG_INF))
 

@@ -112,15 +112,17 @@ def mining_stats(S: np.ndarray, same: np.ndarray, diff: np.ndarray) -> MiningSta
 
 def relative_index(sn: float, size: int) -> int:
     """The reference's relative-position arithmetic (.cu:285-287 etc.):
-    sn >= 0 -> size-1-int(sn); sn < 0 -> int(size-1 + sn*size), computed in
-    double from the float32 param, truncation toward zero; we clamp to the
-    valid range (reference is UB out of range)."""
+    sn >= 0 -> size-1-int(sn); sn < 0 -> int(size-1 + sn*size).  The C
+    expression is `size_t - 1 + float * size_t`, whose usual arithmetic
+    conversions promote everything to FLOAT32 — e.g. sn=-0.3, size=10
+    gives 9 + (-3.0000001f) = 5.9999999f -> 5, not 6.  Truncation toward
+    zero; we clamp to the valid range (reference is UB out of range)."""
     if size <= 0:
         return -1
     if sn >= 0:  # note: -0.0 >= 0 is True, matching C
         pos = size - 1 - int(sn)
     else:
-        pos = int(size - 1 + sn * size)
+        pos = int(np.float32(np.float32(size - 1) + np.float32(sn) * np.float32(size)))
     return min(max(pos, 0), size - 1)
 
 

@@ -173,7 +173,9 @@ def test_relative_index_formula():
     assert oracle.relative_index(0.0, 10) == 9
     assert oracle.relative_index(-0.0, 10) == 9  # -0.0 takes the >= 0 branch
     assert oracle.relative_index(3.0, 10) == 6
-    assert oracle.relative_index(-0.3, 10) == 6  # 9 + (-3.0000...) -> 5.999 -> truncates
+    # float32 arithmetic (the reference's size_t+float promotion):
+    # -0.3f * 10 rounds to exactly -3.0f (round-to-even) -> 9 - 3 = 6
+    assert oracle.relative_index(-0.3, 10) == 6
     assert oracle.relative_index(-1.0, 10) == 0  # clamped (reference UB)
     assert oracle.relative_index(100.0, 10) == 0  # clamped (reference UB)
 
