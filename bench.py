@@ -1,0 +1,176 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GoogLeNet N-pair metric-learning training step.
+
+Measures the BASELINE.json headline metric — images/sec (whole node) for
+GoogLeNet 1024-d N-pair training at batch 256/GPU with the production
+mining config (GLOBAL RELATIVE_HARD ap + LOCAL HARD an), synthetic data,
+random-init weights, bf16 autocast backbone + fp32 loss path.
+
+Single GPU:   python bench.py --gpus 1 --steps 30 --warmup 5
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N ...
+(one rank per GPU over RCCL; reads RANK/LOCAL_RANK/WORLD_SIZE/MASTER_*)
+
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=256, help="per-GPU batch")
+    p.add_argument("--model", type=str, default="googlenet")
+    p.add_argument("--image", type=int, default=224)
+    p.add_argument("--ids-per-batch", type=int, default=None,
+                   help="P identities per batch (default batch//2)")
+    p.add_argument("--no-amp", action="store_true")
+    p.add_argument("--profile-trace", action="store_true",
+                   help="emit a torch profiler trace for rank 0")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    env_ws = int(os.environ.get("WORLD_SIZE", "1"))
+    world = max(env_ws, 1)
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    if world > 1:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        dist.init_process_group("nccl" if use_cuda else "gloo",
+                                rank=rank, world_size=world)
+
+    from npairloss_amd.config.params import NPairLossConfig
+    from npairloss_amd.models import build_embedding_model
+    from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
+    from npairloss_amd.engine.solver import CaffeSGD
+    from npairloss_amd.parallel.ddp import BucketedGradReducer
+
+    torch.manual_seed(1234 + rank)
+    B = args.batch
+    P = args.ids_per_batch or max(2, B // 2)
+    K = B // P
+
+    model = build_embedding_model(args.model).to(device)
+    model = model.to(memory_format=torch.channels_last)
+    model.train()
+    cfg = NPairLossConfig(
+        margin_ident=0.0, margin_diff=-0.05, identsn=-0.0, diffsn=-0.3,
+        ap_mining_region="GLOBAL", ap_mining_method="RELATIVE_HARD",
+        an_mining_region="LOCAL", an_mining_method="HARD")
+    loss_mod = NPairMultiClassLoss(cfg)
+    opt = CaffeSGD(model.parameters(), lr=0.001, momentum=0.9, weight_decay=2e-5)
+    reducer = BucketedGradReducer(model)
+    reducer.broadcast_params()
+
+    amp = use_cuda and not args.no_amp
+
+    # synthetic data: two alternating P x K-labelled image batches on device
+    def make_batch(seed):
+        g = torch.Generator(device="cpu").manual_seed(seed)
+        x = torch.randn(B, 3, args.image, args.image, generator=g)
+        lab = torch.arange(P).repeat_interleave(K)[:B]
+        perm = torch.randperm(B, generator=g)
+        return (x.to(device).to(memory_format=torch.channels_last),
+                lab[perm].to(device))
+    batches = [make_batch(1000 + rank * 10 + i) for i in range(2)]
+
+    last_out = {}
+
+    def step(i):
+        nonlocal last_out
+        x, lab = batches[i % 2]
+        opt.zero_grad(set_to_none=True)
+        if amp:
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                feats = model(x)
+            out = loss_mod(feats.float(), lab)
+        else:
+            feats = model(x)
+            out = loss_mod(feats, lab)
+        out.loss.backward()
+        reducer.finalize()
+        opt.step()
+        last_out = {"loss": out.loss.detach(), "top1": out.retrieve_top1.detach()}
+
+    def barrier_sync():
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        step(i)
+    barrier_sync()
+    t0 = time.time()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    barrier_sync()
+    elapsed = time.time() - t0
+
+    # MAX elapsed over ranks
+    if world > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    images_per_sec = world * B * args.steps / elapsed
+    if rank == 0:
+        result = {
+            "metric": "images/sec (whole node), GoogLeNet N-pair batch=256/GPU",
+            "value": images_per_sec,
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if amp else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": world * B,
+                "batch_per_gpu": B,
+                "image": args.image,
+                "embed_dim": 1024 if args.model == "googlenet" else None,
+                "mining": "ap GLOBAL RELATIVE_HARD + an LOCAL HARD",
+                "parallelism": f"dp{world}",
+                "recall_top1_last_step": float(last_out["top1"]) if last_out else None,
+            },
+        }
+        print(json.dumps(result))
+
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

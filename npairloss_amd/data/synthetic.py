@@ -1,0 +1,63 @@
+"""Synthetic datasets (there is no network access for real datasets; the
+bench contract requires synthetic data of the production shape).
+
+SyntheticImageDataset: class-conditional random images — each class has a
+fixed random texture pattern plus per-sample noise, so metric learning has
+actual signal and the online Recall@k metrics move during smoke training.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+from torch.utils.data import Dataset
+
+
+class SyntheticImageDataset(Dataset):
+    def __init__(self, num_classes: int = 256, per_class: int = 16,
+                 image_size: int = 224, seed: int = 0, noise: float = 0.3):
+        self.num_classes = num_classes
+        self.per_class = per_class
+        self.image_size = image_size
+        self.noise = noise
+        self.labels = np.repeat(np.arange(num_classes), per_class).tolist()
+        self.seed = seed
+        # low-res class patterns, upsampled on access (keeps memory small)
+        g = torch.Generator().manual_seed(seed)
+        self._patterns = torch.randn(num_classes, 3, 8, 8, generator=g)
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, idx):
+        lab = self.labels[idx]
+        g = torch.Generator().manual_seed(self.seed * 1000003 + idx)
+        base = torch.nn.functional.interpolate(
+            self._patterns[lab : lab + 1], size=(self.image_size, self.image_size),
+            mode="bilinear", align_corners=False)[0]
+        img = base + self.noise * torch.randn(3, self.image_size, self.image_size, generator=g)
+        return img, lab
+
+
+class SyntheticEmbeddingDataset(Dataset):
+    """Clustered unit-norm embeddings (backbone-free loss testing)."""
+
+    def __init__(self, num_classes: int = 64, per_class: int = 8, dim: int = 64,
+                 seed: int = 0, noise: float = 0.3):
+        rng = np.random.default_rng(seed)
+        centers = rng.standard_normal((num_classes, dim))
+        centers /= np.linalg.norm(centers, axis=1, keepdims=True)
+        feats, labels = [], []
+        for c in range(num_classes):
+            x = centers[c] + noise * rng.standard_normal((per_class, dim))
+            x /= np.linalg.norm(x, axis=1, keepdims=True)
+            feats.append(x)
+            labels.extend([c] * per_class)
+        self.features = torch.from_numpy(np.concatenate(feats)).float()
+        self.labels = labels
+
+    def __len__(self):
+        return len(self.labels)
+
+    def __getitem__(self, idx):
+        return self.features[idx], self.labels[idx]

@@ -1,0 +1,58 @@
+"""SGD solver with Caffe semantics.
+
+The reference trains with the fork's Caffe solver (usage/solver.prototxt:
+SGD, base_lr 0.001, step x0.5/10k, momentum 0.9, weight_decay 2e-5).
+Caffe's update rule differs from torch.optim.SGD in where lr enters:
+    Caffe:  v = momentum * v + lr * (grad + wd * w);  w -= v
+    torch:  v = momentum * v + (grad + wd * w);       w -= lr * v
+(the two diverge whenever lr changes mid-run, which the step policy does
+every 10k iterations), so CaffeSGD implements the Caffe rule exactly.
+"""
+
+from __future__ import annotations
+
+from typing import Iterable
+
+import torch
+
+from ..config.params import SolverConfig
+
+
+class CaffeSGD(torch.optim.Optimizer):
+    def __init__(self, params: Iterable, lr: float = 0.01, momentum: float = 0.0,
+                 weight_decay: float = 0.0):
+        defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            lr = group["lr"]
+            mom = group["momentum"]
+            wd = group["weight_decay"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                g = p.grad
+                if wd != 0:
+                    g = g.add(p, alpha=wd)
+                st = self.state[p]
+                if "v" not in st:
+                    st["v"] = torch.zeros_like(p)
+                v = st["v"]
+                v.mul_(mom).add_(g, alpha=lr)  # v = mom*v + lr*g
+                p.add_(v, alpha=-1.0)
+        return loss
+
+    def set_lr(self, lr: float):
+        for group in self.param_groups:
+            group["lr"] = lr
+
+
+def build_optimizer(model: torch.nn.Module, solver: SolverConfig) -> CaffeSGD:
+    return CaffeSGD(model.parameters(), lr=solver.base_lr,
+                    momentum=solver.momentum, weight_decay=solver.weight_decay)

@@ -1,0 +1,172 @@
+"""Training loop with Caffe-solver semantics.
+
+Drives the implied trainer of the reference's prototxts: per-iteration
+forward (backbone -> L2Normalize -> NPairMultiClassLoss) and backward,
+Caffe-rule SGD with the step LR policy, `display`-interval logging with
+`average_loss`-window smoothing (solver.prototxt:5-6), periodic TEST-phase
+evaluation (test_interval x test_iter, :2-3), and snapshot/resume
+(:15-16).  Multi-GPU: one process per GPU; the loss module all-gathers
+embeddings internally; backbone grads go through the bucketed RCCL
+all-reduce overlapped with backward (parallel/ddp.py).
+"""
+
+from __future__ import annotations
+
+import collections
+import json
+import os
+import time
+from typing import Callable, Iterable, Optional
+
+import torch
+
+from ..config.params import SolverConfig
+from ..ops.npair_loss import NPairMultiClassLoss
+from ..parallel import collectives as comm
+from ..parallel.ddp import BucketedGradReducer
+from .solver import CaffeSGD, build_optimizer
+
+
+class AverageWindow:
+    """Caffe's average_loss smoothing (running mean over the last N)."""
+
+    def __init__(self, n: int):
+        self.buf = collections.deque(maxlen=max(1, n))
+
+    def add(self, v: float) -> float:
+        self.buf.append(v)
+        return sum(self.buf) / len(self.buf)
+
+
+class Trainer:
+    def __init__(self, model: torch.nn.Module, loss: NPairMultiClassLoss,
+                 solver: SolverConfig, train_loader: Iterable,
+                 test_loader: Optional[Iterable] = None,
+                 device: Optional[torch.device] = None,
+                 amp_dtype: Optional[torch.dtype] = None,
+                 log_fn: Callable[[str], None] = print,
+                 channels_last: bool = False):
+        self.device = device or (torch.device("cuda") if torch.cuda.is_available()
+                                 else torch.device("cpu"))
+        self.model = model.to(self.device)
+        if channels_last:
+            self.model = self.model.to(memory_format=torch.channels_last)
+        self.channels_last = channels_last
+        self.loss = loss
+        self.solver = solver
+        self.train_loader = train_loader
+        self.test_loader = test_loader
+        self.amp_dtype = amp_dtype
+        self.log = log_fn
+        self.optimizer = build_optimizer(self.model, solver)
+        self.reducer = BucketedGradReducer(self.model)
+        self.reducer.broadcast_params()
+        self.iter = 0
+        self.avg = AverageWindow(solver.average_loss)
+        self.history = []
+
+    # -- checkpointing ------------------------------------------------------
+
+    def snapshot(self, prefix: Optional[str] = None):
+        if comm.rank() != 0:
+            return None
+        prefix = prefix or self.solver.snapshot_prefix or "./snap/model_"
+        os.makedirs(os.path.dirname(prefix) or ".", exist_ok=True)
+        path = f"{prefix}iter_{self.iter}.pt"
+        torch.save({
+            "iter": self.iter,
+            "model": self.model.state_dict(),
+            "optimizer": self.optimizer.state_dict(),
+            "solver": self.solver.__dict__,
+        }, path)
+        return path
+
+    def restore(self, path: str):
+        ck = torch.load(path, map_location=self.device, weights_only=False)
+        self.model.load_state_dict(ck["model"])
+        self.optimizer.load_state_dict(ck["optimizer"])
+        self.iter = ck["iter"]
+
+    # -- one training iteration --------------------------------------------
+
+    def train_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
+        self.model.train()
+        images = images.to(self.device, non_blocking=True)
+        if self.channels_last and images.dim() == 4:
+            images = images.to(memory_format=torch.channels_last)
+        labels = labels.to(self.device, non_blocking=True)
+        lr = self.solver.lr_at(self.iter)
+        self.optimizer.set_lr(lr)
+        self.optimizer.zero_grad(set_to_none=True)
+        if self.amp_dtype is not None and self.device.type == "cuda":
+            with torch.autocast("cuda", dtype=self.amp_dtype):
+                feats = self.model(images)
+            out = self.loss(feats.float(), labels)
+        else:
+            feats = self.model(images)
+            out = self.loss(feats, labels)
+        out.loss.backward()
+        self.reducer.finalize()
+        self.optimizer.step()
+        self.iter += 1
+        return {
+            "loss": out.loss, "top1": out.retrieve_top1, "top5": out.retrieve_top5,
+            "top10": out.retrieve_top10, "asum": out.feature_asum, "lr": lr,
+        }
+
+    # -- evaluation ---------------------------------------------------------
+
+    @torch.no_grad()
+    def evaluate(self, max_batches: Optional[int] = None) -> dict:
+        if self.test_loader is None:
+            return {}
+        self.model.eval()
+        n = 0
+        acc = collections.defaultdict(float)
+        for images, labels in self.test_loader:
+            images = images.to(self.device, non_blocking=True)
+            labels = labels.to(self.device, non_blocking=True)
+            feats = self.model(images)
+            out = self.loss(feats, labels)
+            for k, v in (("loss", out.loss), ("top1", out.retrieve_top1),
+                         ("top5", out.retrieve_top5), ("top10", out.retrieve_top10)):
+                acc[k] += float(v)
+            n += 1
+            if max_batches is not None and n >= max_batches:
+                break
+        return {k: v / max(n, 1) for k, v in acc.items()}
+
+    # -- main loop ----------------------------------------------------------
+
+    def fit(self, max_iter: Optional[int] = None):
+        max_iter = max_iter or self.solver.max_iter
+        t0 = time.time()
+        it_timer = time.time()
+        data_iter = iter(self.train_loader)
+        while self.iter < max_iter:
+            try:
+                images, labels = next(data_iter)
+            except StopIteration:
+                data_iter = iter(self.train_loader)
+                images, labels = next(data_iter)
+            stats = self.train_step(images, labels)
+            if self.solver.display and self.iter % self.solver.display == 0:
+                loss_v = float(stats["loss"])
+                sm = self.avg.add(loss_v)
+                dt = time.time() - it_timer
+                it_timer = time.time()
+                if comm.rank() == 0:
+                    self.log(
+                        f"iter {self.iter} lr {stats['lr']:.3g} loss {loss_v:.4f} "
+                        f"(avg {sm:.4f}) top1 {float(stats['top1']):.3f} "
+                        f"top5 {float(stats['top5']):.3f} asum {float(stats['asum']):.3f} "
+                        f"[{dt:.1f}s]")
+                self.history.append({"iter": self.iter, "loss": loss_v, "lr": stats["lr"]})
+            if (self.solver.test_interval and self.test_loader is not None
+                    and self.iter % self.solver.test_interval == 0):
+                ev = self.evaluate(max_batches=self.solver.test_iter or None)
+                if comm.rank() == 0 and ev:
+                    self.log(f"TEST iter {self.iter}: " + json.dumps(ev))
+            if self.solver.snapshot and self.iter % self.solver.snapshot == 0:
+                self.snapshot()
+        return time.time() - t0

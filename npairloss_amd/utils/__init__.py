@@ -1,0 +1,7 @@
+from .caffemodel import (
+    read_caffemodel,
+    write_caffemodel,
+    load_caffemodel_into,
+)
+
+__all__ = ["read_caffemodel", "write_caffemodel", "load_caffemodel_into"]
