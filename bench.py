@@ -53,6 +53,7 @@ def main():
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True  # MIOpen find-best for fixed shapes
 
     if world > 1:
         import torch.distributed as dist

@@ -64,6 +64,7 @@ class Trainer:
         self.iter = 0
         self.avg = AverageWindow(solver.average_loss)
         self.history = []
+        self.augment = None  # optional TRAIN-phase batch augmentation (DataTransformer)
 
     # -- checkpointing ------------------------------------------------------
 
@@ -92,6 +93,8 @@ class Trainer:
     def train_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
         self.model.train()
         images = images.to(self.device, non_blocking=True)
+        if self.augment is not None and images.dim() == 4:
+            images = self.augment(images)
         if self.channels_last and images.dim() == 4:
             images = images.to(memory_format=torch.channels_last)
         labels = labels.to(self.device, non_blocking=True)

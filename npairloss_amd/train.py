@@ -1,0 +1,61 @@
+"""Training CLI: `python -m npairloss_amd.train --solver solver.prototxt`.
+
+Drives the whole reference workflow from the reference's own config files:
+the solver prototxt names the net prototxt (`net:` field), which defines
+the data layer, augmentation, backbone, L2Normalize and the
+NPairMultiClassLoss layer with its mining config.  Multi-GPU: launch with
+torch.distributed.run, one rank per GPU (RCCL).
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+
+import torch
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser()
+    p.add_argument("--solver", required=True, help="solver.prototxt path")
+    p.add_argument("--net", default=None, help="override net prototxt path")
+    p.add_argument("--weights", default=None, help=".caffemodel or .pt to load")
+    p.add_argument("--max-iter", type=int, default=None)
+    p.add_argument("--synthetic-classes", type=int, default=256)
+    p.add_argument("--amp", choices=["off", "bf16", "fp16"], default="bf16")
+    p.add_argument("--num-workers", type=int, default=2)
+    args = p.parse_args(argv)
+
+    from .config.params import SolverConfig
+    from .engine.net_builder import build_trainer_from_prototxt
+
+    solver = SolverConfig.from_prototxt(open(args.solver).read())
+    net_path = args.net or solver.net
+    if net_path and not os.path.isabs(net_path):
+        cand = os.path.join(os.path.dirname(os.path.abspath(args.solver)), net_path)
+        net_path = cand if os.path.exists(cand) else net_path
+    net_text = open(net_path).read()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world > 1:
+        import torch.distributed as dist
+
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo")
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+
+    amp_dtype = {"off": None, "bf16": torch.bfloat16, "fp16": torch.float16}[args.amp]
+    caffemodel = args.weights if (args.weights and args.weights.endswith(".caffemodel")) else None
+    trainer = build_trainer_from_prototxt(
+        net_text, solver, synthetic_classes=args.synthetic_classes,
+        amp_dtype=amp_dtype, caffemodel=caffemodel,
+        num_workers=args.num_workers)
+    if args.weights and args.weights.endswith(".pt"):
+        trainer.restore(args.weights)
+    trainer.fit(max_iter=args.max_iter)
+
+
+if __name__ == "__main__":
+    main()
