@@ -17,6 +17,8 @@ from typing import Dict, Tuple
 import torch
 from torch import nn
 
+from ..ops.vision import CrossChannelLRN, MaxPool3x3
+
 
 class ConvReLU(nn.Module):
     def __init__(self, cin: int, cout: int, k: int, stride: int = 1, pad: int = 0):
@@ -38,7 +40,7 @@ class Inception(nn.Module):
         self.b3 = ConvReLU(c3r, c3, 3, pad=1)
         self.b5_reduce = ConvReLU(cin, c5r, 1)
         self.b5 = ConvReLU(c5r, c5, 5, pad=2)
-        self.pool = nn.MaxPool2d(3, stride=1, padding=1, ceil_mode=True)
+        self.pool = MaxPool3x3(stride=1)  # fused gfx950 kernel on GPU
         self.pool_proj = ConvReLU(cin, cp, 1)
 
     def forward(self, x):
@@ -72,17 +74,17 @@ class GoogLeNet(nn.Module):
     def __init__(self, dropout: float = 0.4):
         super().__init__()
         self.conv1 = ConvReLU(3, 64, 7, stride=2, pad=3)
-        self.pool1 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
-        self.norm1 = nn.LocalResponseNorm(5, alpha=1e-4, beta=0.75)
+        self.pool1 = MaxPool3x3(stride=2)
+        self.norm1 = CrossChannelLRN(5, alpha=1e-4, beta=0.75)
         self.conv2_reduce = ConvReLU(64, 64, 1)
         self.conv2 = ConvReLU(64, 192, 3, pad=1)
-        self.norm2 = nn.LocalResponseNorm(5, alpha=1e-4, beta=0.75)
-        self.pool2 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
+        self.norm2 = CrossChannelLRN(5, alpha=1e-4, beta=0.75)
+        self.pool2 = MaxPool3x3(stride=2)
         self.inception = nn.ModuleDict({
             name: Inception(_INCEPTION_IN[name], *cfg) for name, cfg in _INCEPTION_CFG.items()
         })
-        self.pool3 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
-        self.pool4 = nn.MaxPool2d(3, stride=2, ceil_mode=True)
+        self.pool3 = MaxPool3x3(stride=2)
+        self.pool4 = MaxPool3x3(stride=2)
         self.pool5 = nn.AdaptiveAvgPool2d(1)  # 7x7 avg at 224 input
         self.dropout = nn.Dropout(dropout)
         self._init_weights()

@@ -36,6 +36,14 @@ torch::Tensor global_relative_thr(torch::Tensor S, torch::Tensor lab_l,
 std::vector<torch::Tensor> l2norm_fwd(torch::Tensor x);
 torch::Tensor l2norm_bwd(torch::Tensor y, torch::Tensor inv_norm, torch::Tensor dy);
 
+// vision.hip
+torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, double k);
+torch::Tensor lrn_bwd(torch::Tensor x, torch::Tensor dy, int64_t size, double alpha,
+                      double beta, double k);
+std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ceil_mode);
+torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
+                           int64_t H, int64_t W);
+
 // gemm_f32.hip
 torch::Tensor sim_gemm_nt(torch::Tensor F_l, torch::Tensor F_g);
 torch::Tensor gemm_nn(torch::Tensor A, torch::Tensor B);
@@ -51,6 +59,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("global_relative_thr", &global_relative_thr, "device radix-select order-statistic threshold");
   m.def("l2norm_fwd", &l2norm_fwd, "row L2 normalize forward");
   m.def("l2norm_bwd", &l2norm_bwd, "row L2 normalize backward");
+  m.def("lrn_fwd", &lrn_fwd, "fused across-channel LRN forward");
+  m.def("lrn_bwd", &lrn_bwd, "fused across-channel LRN backward");
+  m.def("maxpool3_fwd", &maxpool3_fwd, "3x3 max pool forward with argmax");
+  m.def("maxpool3_bwd", &maxpool3_bwd, "3x3 max pool gather backward");
   m.def("sim_gemm_nt", &sim_gemm_nt, "fp32 MFMA similarity GEMM (A @ B^T)");
   m.def("gemm_nn", &gemm_nn, "fp32 MFMA GEMM A @ B");
   m.def("gemm_tn", &gemm_tn, "fp32 MFMA GEMM A^T @ B");

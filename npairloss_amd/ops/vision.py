@@ -1,0 +1,97 @@
+"""Vision ops backed by the fused gfx950 kernels (GPU) or torch (CPU).
+
+CrossChannelLRN and MaxPool3x3 replace torch's eager LRN chain and
+atomic-based max-pool backward in the GoogLeNet path — rocprof showed
+those two dominating the flagship bench step (profiles/).
+"""
+
+from __future__ import annotations
+
+import torch
+from torch import nn
+import torch.nn.functional as F
+
+from . import _backend
+
+
+class _LRNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, size, alpha, beta, k):
+        ctx.params = (size, alpha, beta, k)
+        if x.is_cuda:
+            y = _backend.ext().lrn_fwd(x, size, alpha, beta, k)
+            ctx.save_for_backward(x)
+            return y
+        ctx.save_for_backward(x)
+        return F.local_response_norm(x, size, alpha=alpha, beta=beta, k=k)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        size, alpha, beta, k = ctx.params
+        if x.is_cuda:
+            dx = _backend.ext().lrn_bwd(x, dy, size, alpha, beta, k)
+        else:
+            with torch.enable_grad():
+                xr = x.detach().requires_grad_(True)
+                y = F.local_response_norm(xr, size, alpha=alpha, beta=beta, k=k)
+                (dx,) = torch.autograd.grad(y, xr, dy)
+        return dx, None, None, None, None
+
+
+class CrossChannelLRN(nn.Module):
+    """Caffe across-channel LRN: y = x * (k + alpha/n * sum_win x^2)^-beta."""
+
+    def __init__(self, size: int = 5, alpha: float = 1e-4, beta: float = 0.75,
+                 k: float = 1.0):
+        super().__init__()
+        self.size, self.alpha, self.beta, self.k = size, alpha, beta, k
+
+    def forward(self, x):
+        return _LRNFn.apply(x, self.size, self.alpha, self.beta, self.k)
+
+    def extra_repr(self):
+        return f"size={self.size}, alpha={self.alpha}, beta={self.beta}, k={self.k}"
+
+
+class _MaxPool3Fn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, stride, ceil_mode):
+        if x.is_cuda:
+            y, idx = _backend.ext().maxpool3_fwd(x, stride, ceil_mode)
+            ctx.save_for_backward(idx)
+            ctx.meta = (stride, x.shape[2], x.shape[3], True)
+            return y
+        ctx.save_for_backward(x)
+        ctx.meta = (stride, x.shape[2], x.shape[3], False)
+        ctx.ceil_mode = ceil_mode
+        return F.max_pool2d(x, 3, stride=stride, padding=1, ceil_mode=ceil_mode)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (saved,) = ctx.saved_tensors
+        stride, H, W, gpu = ctx.meta
+        if gpu:
+            dx = _backend.ext().maxpool3_bwd(dy, saved, stride, H, W)
+        else:
+            with torch.enable_grad():
+                xr = saved.detach().requires_grad_(True)
+                y = F.max_pool2d(xr, 3, stride=stride, padding=1, ceil_mode=ctx.ceil_mode)
+                (dx,) = torch.autograd.grad(y, xr, dy)
+        return dx, None, None
+
+
+class MaxPool3x3(nn.Module):
+    """3x3 max pool, padding 1, stride 1 or 2, Caffe ceil_mode."""
+
+    def __init__(self, stride: int = 1, ceil_mode: bool = True):
+        super().__init__()
+        assert stride in (1, 2)
+        self.stride = stride
+        self.ceil_mode = ceil_mode
+
+    def forward(self, x):
+        return _MaxPool3Fn.apply(x, self.stride, self.ceil_mode)
+
+    def extra_repr(self):
+        return f"stride={self.stride}, ceil_mode={self.ceil_mode}"

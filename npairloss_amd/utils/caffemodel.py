@@ -220,6 +220,24 @@ def write_caffemodel(path: str, layers: List[CaffeLayer], net_name: str = "net")
 # ---------------------------------------------------------------------------
 
 
+def save_caffemodel(model: torch.nn.Module, path: str, net_name: str = "net") -> int:
+    """Write the model's caffe-named weights as a .caffemodel (the inverse
+    of load_caffemodel_into; Caffe itself can read the result)."""
+    target = model
+    if not hasattr(target, "caffe_names") and hasattr(target, "backbone"):
+        target = target.backbone
+    if not hasattr(target, "caffe_names"):
+        raise TypeError("model does not expose caffe_names()")
+    layers = []
+    for name, mod in target.caffe_names().items():
+        blobs = [mod.weight.detach().cpu().numpy()]
+        if getattr(mod, "bias", None) is not None:
+            blobs.append(mod.bias.detach().cpu().numpy())
+        layers.append(CaffeLayer(name, "Convolution", blobs))
+    write_caffemodel(path, layers, net_name=net_name)
+    return len(layers)
+
+
 def load_caffemodel_into(model: torch.nn.Module, path_or_bytes,
                          strict: bool = False) -> Tuple[List[str], List[str]]:
     """Copy blobs into `model` by Caffe layer name.

@@ -29,6 +29,7 @@ SOURCES = [
     "sort_select.hip",
     "l2norm.hip",
     "gemm_f32.hip",
+    "vision.hip",
 ]
 
 

@@ -174,3 +174,18 @@ def test_synthetic_image_dataset():
     c, lc = ds[2]
     assert la == lb and la != lc
     assert (a - b).abs().mean() < (a - c).abs().mean()
+
+
+def test_caffemodel_save_load_roundtrip(tmp_path):
+    from npairloss_amd.utils.caffemodel import save_caffemodel
+    m1 = GoogLeNet()
+    p = str(tmp_path / "snap.caffemodel")
+    n = save_caffemodel(m1, p)
+    assert n == len(m1.caffe_names())
+    m2 = GoogLeNet()
+    loaded, skipped = load_caffemodel_into(m2, p, strict=True)
+    assert not skipped
+    for name, conv in m1.caffe_names().items():
+        conv2 = m2.caffe_names()[name]
+        torch.testing.assert_close(conv.weight, conv2.weight)
+        torch.testing.assert_close(conv.bias, conv2.bias)
