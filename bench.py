@@ -38,8 +38,8 @@ def parse_args():
     p.add_argument("--ids-per-batch", type=int, default=None,
                    help="P identities per batch (default batch//2)")
     p.add_argument("--no-amp", action="store_true")
-    p.add_argument("--profile-trace", action="store_true",
-                   help="emit a torch profiler trace for rank 0")
+    p.add_argument("--timers", action="store_true",
+                   help="per-phase HIP-event timing report on stderr (rank 0)")
     return p.parse_args()
 
 
@@ -99,21 +99,27 @@ def main():
     batches = [make_batch(1000 + rank * 10 + i) for i in range(2)]
 
     last_out = {}
+    from npairloss_amd.utils.profiling import PhaseTimers
+    timers = PhaseTimers(enabled=args.timers, use_cuda=use_cuda)
 
     def step(i):
         nonlocal last_out
         x, lab = batches[i % 2]
         opt.zero_grad(set_to_none=True)
-        if amp:
-            with torch.autocast("cuda", dtype=torch.bfloat16):
+        with timers.phase("forward"):
+            if amp:
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    feats = model(x)
+            else:
                 feats = model(x)
+        with timers.phase("loss"):
             out = loss_mod(feats.float(), lab)
-        else:
-            feats = model(x)
-            out = loss_mod(feats, lab)
-        out.loss.backward()
-        reducer.finalize()
-        opt.step()
+        with timers.phase("backward"):
+            out.loss.backward()
+        with timers.phase("comm"):
+            reducer.finalize()
+        with timers.phase("optimizer"):
+            opt.step()
         last_out = {"loss": out.loss.detach(), "top1": out.retrieve_top1.detach()}
 
     def barrier_sync():
@@ -125,6 +131,7 @@ def main():
 
     for i in range(args.warmup):
         step(i)
+    timers.report(reset=True)  # drop warmup phases
     barrier_sync()
     t0 = time.time()
     for i in range(args.steps):
@@ -141,6 +148,11 @@ def main():
         elapsed = float(t.item())
 
     images_per_sec = world * B * args.steps / elapsed
+    if args.timers and rank == 0:
+        ph = timers.report()
+        per_step = {k: v / args.steps for k, v in sorted(ph.items(), key=lambda kv: -kv[1])}
+        print("phases ms/step: " + json.dumps({k: round(v, 2) for k, v in per_step.items()}),
+              file=sys.stderr)
     if rank == 0:
         result = {
             "metric": "images/sec (whole node), GoogLeNet N-pair batch=256/GPU",

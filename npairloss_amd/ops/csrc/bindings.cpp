@@ -44,6 +44,11 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
 torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
                            int64_t H, int64_t W);
 
+// gemm_lowp.hip
+torch::Tensor sim_gemm_nt_bf16(torch::Tensor A, torch::Tensor B);
+torch::Tensor sim_gemm_nt_fp8(torch::Tensor A, torch::Tensor B);
+torch::Tensor cast_fp8(torch::Tensor x);
+
 // gemm_f32.hip
 torch::Tensor sim_gemm_nt(torch::Tensor F_l, torch::Tensor F_g);
 torch::Tensor gemm_nn(torch::Tensor A, torch::Tensor B);
@@ -64,6 +69,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3_fwd", &maxpool3_fwd, "3x3 max pool forward with argmax");
   m.def("maxpool3_bwd", &maxpool3_bwd, "3x3 max pool gather backward");
   m.def("sim_gemm_nt", &sim_gemm_nt, "fp32 MFMA similarity GEMM (A @ B^T)");
+  m.def("sim_gemm_nt_bf16", &sim_gemm_nt_bf16, "bf16 MFMA similarity GEMM");
+  m.def("sim_gemm_nt_fp8", &sim_gemm_nt_fp8, "fp8 e4m3 MFMA similarity GEMM");
+  m.def("cast_fp8", &cast_fp8, "fp32 -> fp8 e4m3 bit pattern");
   m.def("gemm_nn", &gemm_nn, "fp32 MFMA GEMM A @ B");
   m.def("gemm_tn", &gemm_tn, "fp32 MFMA GEMM A^T @ B");
 }

@@ -220,10 +220,18 @@ def _bwd_weights_torch(S, labels_l, labels_g, rank, thr_p, thr_n, max_all,
 # GPU (HIP extension) implementation
 # ---------------------------------------------------------------------------
 
-def _forward_hip(F_l, labels_l, F_g, labels_g, rank, cfg: NPairLossConfig, ks):
+def _forward_hip(F_l, labels_l, F_g, labels_g, rank, cfg: NPairLossConfig, ks,
+                 sim_dtype: str = "fp32"):
     C = _backend.ext()
     B = F_l.shape[0]
-    S = C.sim_gemm_nt(F_l, F_g)
+    # similarity GEMM precision: fp32 MFMA (exact, default), bf16 MFMA, or
+    # fp8-e4m3 MFMA (unit-norm embeddings fit e4m3 without scaling)
+    if sim_dtype == "bf16":
+        S = C.sim_gemm_nt_bf16(F_l.to(torch.bfloat16), F_g.to(torch.bfloat16))
+    elif sim_dtype == "fp8":
+        S = C.sim_gemm_nt_fp8(C.cast_fp8(F_l), C.cast_fp8(F_g))
+    else:
+        S = C.sim_gemm_nt(F_l, F_g)
     lab_l = labels_l.to(torch.int32)
     lab_g = labels_g.to(torch.int32)
     min_within, max_between, max_all = C.rowstats(S, lab_l, lab_g, rank)
@@ -268,7 +276,7 @@ def _forward_hip(F_l, labels_l, F_g, labels_g, rank, cfg: NPairLossConfig, ks):
 class _NPairLossFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, features: torch.Tensor, labels: torch.Tensor, cfg: NPairLossConfig,
-                ks: Tuple[int, ...], group):
+                ks: Tuple[int, ...], group, sim_dtype: str = "fp32"):
         in_dtype = features.dtype
         F_l = features.detach().float().contiguous()
         labels = labels.detach()
@@ -278,7 +286,8 @@ class _NPairLossFn(torch.autograd.Function):
         labels_g = comm.all_gather_rows(labels.contiguous(), group)
 
         if F_l.is_cuda:
-            loss, recalls, saved = _forward_hip(F_l, labels, F_g, labels_g, rank, cfg, ks)
+            loss, recalls, saved = _forward_hip(F_l, labels, F_g, labels_g, rank, cfg, ks,
+                                                sim_dtype)
         else:
             loss, recalls, saved = _forward_torch(F_l, labels, F_g, labels_g, rank, cfg, ks)
 
@@ -321,7 +330,7 @@ class _NPairLossFn(torch.autograd.Function):
         # then 0.5*local + 0.5*slice (.cu:462-497) == reduce-scatter form:
         slice_sum = comm.reduce_scatter_rows(dF_t, ctx.group)
         grad = 0.5 * dF_l + (0.5 / ctx.ws) * slice_sum
-        return grad.to(ctx.in_dtype), None, None, None, None
+        return grad.to(ctx.in_dtype), None, None, None, None, None
 
 
 class NPairLossOutput(NamedTuple):
@@ -345,16 +354,20 @@ class NPairMultiClassLoss(nn.Module):
     """
 
     def __init__(self, cfg: Optional[NPairLossConfig] = None,
-                 top_k: Tuple[int, ...] = (1, 5, 10), group=None):
+                 top_k: Tuple[int, ...] = (1, 5, 10), group=None,
+                 sim_dtype: str = "fp32"):
         super().__init__()
+        assert sim_dtype in ("fp32", "bf16", "fp8")
         self.cfg = cfg if cfg is not None else NPairLossConfig()
         self.top_k = tuple(top_k)
         self.group = group
+        self.sim_dtype = sim_dtype  # GPU similarity-GEMM precision
 
     def forward(self, features: torch.Tensor, labels: torch.Tensor) -> NPairLossOutput:
         if features.dim() != 2:
             features = features.flatten(1)  # B x C x H x W -> B x D like Caffe
-        loss, recalls, asum = _NPairLossFn.apply(features, labels, self.cfg, self.top_k, self.group)
+        loss, recalls, asum = _NPairLossFn.apply(features, labels, self.cfg, self.top_k,
+                                                 self.group, self.sim_dtype)
         r = [recalls[i] for i in range(len(self.top_k))]
         while len(r) < 3:
             r.append(torch.zeros_like(loss))

@@ -30,6 +30,7 @@ SOURCES = [
     "l2norm.hip",
     "gemm_f32.hip",
     "vision.hip",
+    "gemm_lowp.hip",
 ]
 
 
