@@ -1,6 +1,7 @@
 from .sampler import PKBatchSampler
 from .synthetic import SyntheticImageDataset, SyntheticEmbeddingDataset
 from .transforms import DataTransformer, TransformConfig
+from .folder import FolderListDataset
 
 __all__ = [
     "PKBatchSampler",
@@ -8,4 +9,5 @@ __all__ = [
     "SyntheticEmbeddingDataset",
     "DataTransformer",
     "TransformConfig",
+    "FolderListDataset",
 ]

@@ -115,15 +115,18 @@ def build_trainer_from_prototxt(
         lcfg = NPairLossConfig()
     loss_mod = NPairMultiClassLoss(lcfg)
 
-    # --- dataset: real image folder if it exists, else synthetic
+    # --- dataset: real list-file folder when it exists, else synthetic
     use_real = spec.root_folder and os.path.isdir(spec.root_folder) and os.path.isfile(spec.source)
     if use_real:
-        raise NotImplementedError(
-            "image-folder MultibatchData source found; offline builds use the "
-            "synthetic dataset (pass synthetic_classes/...)")
-    n_cls = max(synthetic_classes, spec.identities_per_batch)
-    ds = SyntheticImageDataset(num_classes=n_cls, per_class=max(synthetic_per_class, spec.imgs_per_identity),
-                               image_size=img, seed=0)
+        from ..data.folder import FolderListDataset
+
+        ds = FolderListDataset(spec.root_folder, spec.source,
+                               new_height=spec.new_height, new_width=spec.new_width)
+    else:
+        n_cls = max(synthetic_classes, spec.identities_per_batch)
+        ds = SyntheticImageDataset(num_classes=n_cls,
+                                   per_class=max(synthetic_per_class, spec.imgs_per_identity),
+                                   image_size=img, seed=0)
     sampler = PKBatchSampler(ds.labels, spec.identities_per_batch, spec.imgs_per_identity,
                              shuffle=spec.shuffle, rand_identity=spec.rand_identity, seed=0)
     loader = torch.utils.data.DataLoader(ds, batch_sampler=sampler, num_workers=num_workers)

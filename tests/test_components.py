@@ -189,3 +189,29 @@ def test_caffemodel_save_load_roundtrip(tmp_path):
         conv2 = m2.caffe_names()[name]
         torch.testing.assert_close(conv.weight, conv2.weight)
         torch.testing.assert_close(conv.bias, conv2.bias)
+
+
+def test_folder_list_dataset(tmp_path):
+    import numpy as np
+    from npairloss_amd.data.folder import FolderListDataset
+
+    root = tmp_path / "imgs"
+    root.mkdir()
+    # npy image HWC uint8
+    np.save(root / "a.npy", (np.random.rand(10, 12, 3) * 255).astype(np.uint8))
+    # pt image CHW float
+    torch.save(torch.rand(3, 8, 8), root / "b.pt")
+    # ppm P6
+    w, h = 6, 4
+    with open(root / "c.ppm", "wb") as fh:
+        fh.write(b"P6\n# comment\n%d %d\n255\n" % (w, h))
+        fh.write((np.arange(w * h * 3) % 256).astype(np.uint8).tobytes())
+    src = tmp_path / "list.txt"
+    src.write_text("a.npy 0\nb.pt 1\nc.ppm 2\n")
+    ds = FolderListDataset(str(root), str(src), new_height=16, new_width=16)
+    assert len(ds) == 3
+    assert ds.labels == [0, 1, 2]
+    for i in range(3):
+        img, lab = ds[i]
+        assert img.shape == (3, 16, 16)
+        assert lab == i
