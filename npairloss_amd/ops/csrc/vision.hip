@@ -6,15 +6,20 @@
 // them with:
 //   lrn_fwd / lrn_bwd       — Caffe across-channel LRN
 //                             (scale_i = k + alpha/n * sum_{win} x_j^2,
-//                              y = x * scale^-beta) as ONE stencil pass per
-//                             direction, bf16 or fp32 in/out, fp32 math.
+//                              y = x * scale^-beta).  The channel window
+//                             is shared through LDS per pixel group, so
+//                             each element costs ONE powf + ONE div (the
+//                             per-element stencil variant costs 5 of each;
+//                             torch's chain costs ~8 full-tensor passes).
 //   maxpool3x3_fwd / _bwd   — kernel-3 max pool (stride 1 or 2, pad 1)
 //                             storing a 1-byte argmax; backward is a
 //                             deterministic GATHER over the <=9 covering
 //                             windows per input element (no atomics).
+//                             NHWC channels are processed 4 wide
+//                             (8-byte bf16 loads) when C % 4 == 0.
 // Both kernels address NHWC (channels_last, the training layout) or NCHW
-// via a channel-stride parameter; lanes walk the contiguous dimension so
-// every access is coalesced.
+// via strides; lanes walk the contiguous dimension so every access is
+// coalesced.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -34,23 +39,101 @@ template <> DEVINL void stf<__hip_bfloat16>(__hip_bfloat16* p, long long i, floa
   p[i] = __float2bfloat16(v);
 }
 
-// ---------------------------------------------------------------------------
-// LRN (across channels, Caffe semantics; local_size n, alpha, beta, k)
-// ---------------------------------------------------------------------------
-// Element (b, c, s) where s indexes the HW plane:
-//   NHWC: idx = (b*S + s)*C + c      (cstride = 1,      "row" base contiguous in c)
-//   NCHW: idx = (b*C + c)*S + s      (cstride = S)
-// One thread per element; the 5-tap window walks c at cstride (L1-cached).
+// 4-wide channel loads (NHWC fast path)
+template <typename T> DEVINL void ld4(const T* p, float* o);
+template <> DEVINL void ld4<float>(const float* p, float* o) {
+  const float4 v = *reinterpret_cast<const float4*>(p);
+  o[0] = v.x; o[1] = v.y; o[2] = v.z; o[3] = v.w;
+}
+template <> DEVINL void ld4<__hip_bfloat16>(const __hip_bfloat16* p, float* o) {
+  const ushort4 v = *reinterpret_cast<const ushort4*>(p);
+  o[0] = __bfloat162float(__hip_bfloat16_raw{v.x});
+  o[1] = __bfloat162float(__hip_bfloat16_raw{v.y});
+  o[2] = __bfloat162float(__hip_bfloat16_raw{v.z});
+  o[3] = __bfloat162float(__hip_bfloat16_raw{v.w});
+}
+template <typename T> DEVINL void st4(T* p, const float* v);
+template <> DEVINL void st4<float>(float* p, const float* v) {
+  *reinterpret_cast<float4*>(p) = make_float4(v[0], v[1], v[2], v[3]);
+}
+template <> DEVINL void st4<__hip_bfloat16>(__hip_bfloat16* p, const float* v) {
+  ushort4 u;
+  u.x = static_cast<__hip_bfloat16_raw>(__float2bfloat16(v[0])).x;
+  u.y = static_cast<__hip_bfloat16_raw>(__float2bfloat16(v[1])).x;
+  u.z = static_cast<__hip_bfloat16_raw>(__float2bfloat16(v[2])).x;
+  u.w = static_cast<__hip_bfloat16_raw>(__float2bfloat16(v[3])).x;
+  *reinterpret_cast<ushort4*>(p) = u;
+}
 
+// ---------------------------------------------------------------------------
+// LRN — LDS-stencil variant (C <= 256) + generic per-element fallback
+// ---------------------------------------------------------------------------
+// "Pixel" = one (b, h, w) site (NHWC: its C channels contiguous) or one
+// (b, s) site for NCHW (channel stride S).  A 256-thread block processes
+// 256/C_pad pixels at once: xsq (and the dy*y/scale cross terms in the
+// backward) land in LDS once, each window sum reads 5 LDS floats.
+
+template <typename T, bool BWD>
+__global__ void lrn_tile_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                T* __restrict__ out, long long npix, int C, int C_pad,
+                                long long cstride, long long pix_to_base_mul,
+                                long long S, int n, float alpha_over_n, float beta,
+                                float k) {
+  __shared__ float xsq[NPAIR_BLOCK];
+  __shared__ float tbuf[NPAIR_BLOCK];
+  const int half = n / 2;
+  const int ppb = NPAIR_BLOCK / C_pad;        // pixels per block
+  const int pl = threadIdx.x / C_pad;         // pixel slot in block
+  const int c = threadIdx.x % C_pad;
+  const int lds0 = pl * C_pad;
+  const bool chan_ok = (c < C) && (pl < ppb);
+  const long long tiles = (npix + ppb - 1) / ppb;
+  for (long long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
+    const long long pix = tile * ppb + pl;
+    const bool act = chan_ok && pix < npix;
+    long long base = 0, idx = 0;
+    float xi = 0.f, gi = 0.f;
+    if (act) {
+      if (cstride == 1) {
+        base = pix * C;                       // NHWC
+      } else {
+        base = (pix / S) * (C * S) + (pix % S);  // NCHW: pix = (b, s)
+      }
+      idx = base + (long long)c * cstride;
+      xi = ldf(x, idx);
+      if (BWD) gi = ldf(dy, idx);
+    }
+    xsq[threadIdx.x] = act ? xi * xi : 0.f;
+    __syncthreads();
+    const int lo = max(0, c - half);
+    const int hi = min(C - 1, c + half);
+    float ss = 0.f;
+    for (int j = lo; j <= hi; ++j) ss += xsq[lds0 + j];
+    const float scale = k + alpha_over_n * ss;
+    const float p = __powf(scale, -beta);     // ONE powf per element
+    if (!BWD) {
+      if (act) stf(out, idx, xi * p);
+      __syncthreads();
+      continue;
+    }
+    const float yi = xi * p;
+    tbuf[threadIdx.x] = act ? gi * yi / scale : 0.f;
+    __syncthreads();
+    float cross = 0.f;
+    for (int j = lo; j <= hi; ++j) cross += tbuf[lds0 + j];
+    if (act) stf(out, idx, gi * p - 2.f * alpha_over_n * beta * xi * cross);
+    __syncthreads();
+  }
+}
+
+// generic per-element fallback (any C): used when C > 256
 template <typename T>
-__global__ void lrn_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
-                               long long total, int C, long long cstride, int n, float alpha_over_n,
-                               float beta, float k) {
+__global__ void lrn_fwd_generic(const T* __restrict__ x, T* __restrict__ y,
+                                long long total, int C, long long cstride, int n,
+                                float alpha_over_n, float beta, float k) {
   const int half = n / 2;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long long)gridDim.x * blockDim.x) {
-    // decompose i into (outer, c, inner) without knowing layout: we pass
-    // indices so that c = (i / cstride) % C  holds for both layouts.
     const long long c = (i / cstride) % C;
     const long long base = i - c * cstride;
     const int lo = max((long long)0, c - half);
@@ -60,30 +143,24 @@ __global__ void lrn_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
       const float v = ldf(x, base + (long long)j * cstride);
       ss += v * v;
     }
-    const float scale = k + alpha_over_n * ss;
-    stf(y, i, ldf(x, i) * __powf(scale, -beta));
+    stf(y, i, ldf(x, i) * __powf(k + alpha_over_n * ss, -beta));
   }
 }
 
-// dx_i = dy_i*scale_i^-beta - 2*alpha/n*beta * x_i * sum_{j in win(i)} dy_j*y_j/scale_j
-// (y and scale recomputed from x — nothing stored between passes)
 template <typename T>
-__global__ void lrn_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy,
-                               T* __restrict__ dx, long long total, int C,
-                               long long cstride, int n, float alpha_over_n,
-                               float beta, float k) {
+__global__ void lrn_bwd_generic(const T* __restrict__ x, const T* __restrict__ dy,
+                                T* __restrict__ dx, long long total, int C,
+                                long long cstride, int n, float alpha_over_n,
+                                float beta, float k) {
   const int half = n / 2;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long long)gridDim.x * blockDim.x) {
     const long long c = (i / cstride) % C;
     const long long base = i - c * cstride;
-    // window for the cross-term: channels j whose window contains c
     const int lo = max((long long)0, c - half);
     const int hi = min((long long)C - 1, c + half);
-    float cross = 0.f;
-    float scale_i = 0.f;
+    float cross = 0.f, scale_i = 0.f;
     for (int j = lo; j <= hi; ++j) {
-      // scale_j = k + a/n * sum_{m in win(j)} x_m^2
       const int jlo = max(0, j - half);
       const int jhi = min(C - 1, j + half);
       float ss = 0.f;
@@ -97,42 +174,41 @@ __global__ void lrn_bwd_kernel(const T* __restrict__ x, const T* __restrict__ dy
       cross += ldf(dy, base + (long long)j * cstride) * yj / scale_j;
     }
     const float xi = ldf(x, i);
-    const float g = ldf(dy, i) * __powf(scale_i, -beta)
-                  - 2.f * alpha_over_n * beta * xi * cross;
-    stf(dx, i, g);
+    stf(dx, i, ldf(dy, i) * __powf(scale_i, -beta) - 2.f * alpha_over_n * beta * xi * cross);
   }
 }
 
 // ---------------------------------------------------------------------------
-// 3x3 max pool, pad 1, stride 1 or 2
+// 3x3 max pool, pad 1, stride 1 or 2 — V channels per thread (V = 4 or 1)
 // ---------------------------------------------------------------------------
-// NHWC addressing: idx(b,h,w,c) = ((b*H + h)*W + w)*C + c
-// NCHW addressing: idx(b,h,w,c) = ((b*C + c)*H + h)*W + w
-// We pass strides (sb, sh, sw, sc) so one kernel serves both; the launch
-// maps threads over the contiguous dim for coalescing.
 
-template <typename T>
+template <typename T, int V>
 __global__ void maxpool3_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                     unsigned char* __restrict__ idx,
-                                    int B, int C, int H, int W, int OH, int OW,
+                                    int B, int Cv, int H, int W, int OH, int OW,
                                     int stride,
                                     long long xsb, long long xsh, long long xsw, long long xsc,
                                     long long ysb, long long ysh, long long ysw, long long ysc) {
-  const long long total = (long long)B * OH * OW * C;
+  // Cv = C / V; i enumerates ((b*OH + oh)*OW + ow)*Cv + cv  (cv fastest)
+  const long long total = (long long)B * OH * OW * Cv;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long long)gridDim.x * blockDim.x) {
-    // i enumerated as ((b*OH + oh)*OW + ow)*C + c  (c fastest — NHWC-friendly)
-    const int c = i % C;
-    long long r = i / C;
+    const int cv = i % Cv;
+    long long r = i / Cv;
     const int ow = r % OW;
     r /= OW;
     const int oh = r % OH;
     const int b = r / OH;
     const int h0 = oh * stride - 1;
     const int w0 = ow * stride - 1;
-    float best = -FLT_MAX;
-    int besti = 0;
-    const long long xb = (long long)b * xsb + (long long)c * xsc;
+    float best[V];
+    int besti[V];
+#pragma unroll
+    for (int v = 0; v < V; ++v) {
+      best[v] = -FLT_MAX;
+      besti[v] = 0;
+    }
+    const long long xb = (long long)b * xsb + (long long)(cv * V) * xsc;
 #pragma unroll
     for (int dh = 0; dh < 3; ++dh) {
       const int h = h0 + dh;
@@ -141,38 +217,56 @@ __global__ void maxpool3_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
       for (int dw = 0; dw < 3; ++dw) {
         const int w = w0 + dw;
         if (w < 0 || w >= W) continue;
-        const float v = ldf(x, xb + (long long)h * xsh + (long long)w * xsw);
-        if (v > best) {
-          best = v;
-          besti = dh * 3 + dw;
+        const long long a = xb + (long long)h * xsh + (long long)w * xsw;
+        float vals[V];
+        if (V == 4) {
+          ld4(x + a, vals);
+        } else {
+          vals[0] = ldf(x, a);
         }
+#pragma unroll
+        for (int v = 0; v < V; ++v)
+          if (vals[v] > best[v]) {
+            best[v] = vals[v];
+            besti[v] = dh * 3 + dw;
+          }
       }
     }
-    const long long yi = (long long)b * ysb + (long long)oh * ysh + (long long)ow * ysw + (long long)c * ysc;
-    stf(y, yi, best);
-    idx[yi] = (unsigned char)besti;
+    const long long yi = (long long)b * ysb + (long long)oh * ysh + (long long)ow * ysw
+                       + (long long)(cv * V) * ysc;
+    if (V == 4) {
+      st4(y + yi, best);
+      uchar4 u;
+      u.x = besti[0]; u.y = besti[1]; u.z = besti[2]; u.w = besti[3];
+      *reinterpret_cast<uchar4*>(idx + yi) = u;
+    } else {
+      stf(y, yi, best[0]);
+      idx[yi] = (unsigned char)besti[0];
+    }
   }
 }
 
-template <typename T>
+template <typename T, int V>
 __global__ void maxpool3_bwd_kernel(const T* __restrict__ dy,
                                     const unsigned char* __restrict__ idx,
                                     T* __restrict__ dx,
-                                    int B, int C, int H, int W, int OH, int OW,
+                                    int B, int Cv, int H, int W, int OH, int OW,
                                     int stride,
                                     long long xsb, long long xsh, long long xsw, long long xsc,
                                     long long ysb, long long ysh, long long ysw, long long ysc) {
-  const long long total = (long long)B * H * W * C;
+  const long long total = (long long)B * H * W * Cv;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long long)gridDim.x * blockDim.x) {
-    const int c = i % C;
-    long long r = i / C;
+    const int cv = i % Cv;
+    long long r = i / Cv;
     const int w = r % W;
     r /= W;
     const int h = r % H;
     const int b = r / H;
-    float acc = 0.f;
-    const long long yb = (long long)b * ysb + (long long)c * ysc;
+    float acc[V];
+#pragma unroll
+    for (int v = 0; v < V; ++v) acc[v] = 0.f;
+    const long long yb = (long long)b * ysb + (long long)(cv * V) * ysc;
     // output windows covering (h, w): oh*stride - 1 <= h <= oh*stride + 1,
     // i.e. oh in [ceil((h-1)/s), floor((h+1)/s)] (negative lower clamps to 0)
     const int oh_lo = max(0, (h - 1 + stride - 1) / stride);
@@ -186,10 +280,29 @@ __global__ void maxpool3_bwd_kernel(const T* __restrict__ dy,
         const int dw = w - (ow * stride - 1);
         if (dw < 0 || dw > 2) continue;
         const long long yi = yb + (long long)oh * ysh + (long long)ow * ysw;
-        if (idx[yi] == (unsigned char)(dh * 3 + dw)) acc += ldf(dy, yi);
+        const unsigned char want = (unsigned char)(dh * 3 + dw);
+        if (V == 4) {
+          const uchar4 u = *reinterpret_cast<const uchar4*>(idx + yi);
+          if (u.x == want || u.y == want || u.z == want || u.w == want) {
+            float g[4];
+            ld4(dy + yi, g);
+            if (u.x == want) acc[0] += g[0];
+            if (u.y == want) acc[1] += g[1];
+            if (u.z == want) acc[2] += g[2];
+            if (u.w == want) acc[3] += g[3];
+          }
+        } else {
+          if (idx[yi] == want) acc[0] += ldf(dy, yi);
+        }
       }
     }
-    stf(dx, (long long)b * xsb + (long long)h * xsh + (long long)w * xsw + (long long)c * xsc, acc);
+    const long long xi = (long long)b * xsb + (long long)h * xsh + (long long)w * xsw
+                       + (long long)(cv * V) * xsc;
+    if (V == 4) {
+      st4(dx + xi, acc);
+    } else {
+      stf(dx, xi, acc[0]);
+    }
   }
 }
 
@@ -205,6 +318,16 @@ static int grid_for(long long total) {
   return (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 4096);
 }
 
+#define VISION_DISPATCH(TENSOR, NAME, ...)                                        \
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half, \
+      (TENSOR).scalar_type(), NAME, [&] {                                         \
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>,          \
+                                 __hip_bfloat16, float>;                          \
+    TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> ||                        \
+                 std::is_same_v<scalar_t, float>), NAME ": bf16/fp32 only");      \
+    __VA_ARGS__                                                                   \
+  })
+
 torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, double k) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4);
   const bool nhwc = is_nhwc(x);
@@ -215,15 +338,24 @@ torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, 
   const long long cstride = nhwc ? 1 : S;
   auto stream = at::hip::getCurrentHIPStream();
   const float aon = (float)(alpha / size);
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
-      x.scalar_type(), "lrn_fwd", [&] {
-    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
-    TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
-                "lrn: bf16/fp32 only");
-    lrn_fwd_kernel<T><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
-        total, (int)C, cstride, (int)size, aon, (float)beta, (float)k);
-  });
+  if (C <= NPAIR_BLOCK) {
+    const int C_pad = (int)((C + 63) / 64) * 64;
+    const long long npix = B * S;
+    const int ppb = NPAIR_BLOCK / C_pad;
+    const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
+    VISION_DISPATCH(x, "lrn_fwd", {
+      lrn_tile_kernel<T, false><<<blocks, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), nullptr,
+          reinterpret_cast<T*>(y.data_ptr()), npix, (int)C, C_pad, cstride, 0, S,
+          (int)size, aon, (float)beta, (float)k);
+    });
+  } else {
+    VISION_DISPATCH(x, "lrn_fwd", {
+      lrn_fwd_generic<T><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
+          total, (int)C, cstride, (int)size, aon, (float)beta, (float)k);
+    });
+  }
   HIP_CHECK_LAST();
   return y;
 }
@@ -241,17 +373,27 @@ torch::Tensor lrn_bwd(torch::Tensor x, torch::Tensor dy, int64_t size, double al
   const long long cstride = nhwc ? 1 : S;
   auto stream = at::hip::getCurrentHIPStream();
   const float aon = (float)(alpha / size);
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
-      x.scalar_type(), "lrn_bwd", [&] {
-    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
-    TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
-                "lrn: bf16/fp32 only");
-    lrn_bwd_kernel<T><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const T*>(xc.data_ptr()),
-        reinterpret_cast<const T*>(dyc.data_ptr()),
-        reinterpret_cast<T*>(dx.data_ptr()),
-        total, (int)C, cstride, (int)size, aon, (float)beta, (float)k);
-  });
+  if (C <= NPAIR_BLOCK) {
+    const int C_pad = (int)((C + 63) / 64) * 64;
+    const long long npix = B * S;
+    const int ppb = NPAIR_BLOCK / C_pad;
+    const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
+    VISION_DISPATCH(x, "lrn_bwd", {
+      lrn_tile_kernel<T, true><<<blocks, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()),
+          reinterpret_cast<const T*>(dyc.data_ptr()),
+          reinterpret_cast<T*>(dx.data_ptr()), npix, (int)C, C_pad, cstride, 0, S,
+          (int)size, aon, (float)beta, (float)k);
+    });
+  } else {
+    VISION_DISPATCH(x, "lrn_bwd", {
+      lrn_bwd_generic<T><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()),
+          reinterpret_cast<const T*>(dyc.data_ptr()),
+          reinterpret_cast<T*>(dx.data_ptr()),
+          total, (int)C, cstride, (int)size, aon, (float)beta, (float)k);
+    });
+  }
   HIP_CHECK_LAST();
   return dx;
 }
@@ -272,7 +414,6 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
   const bool nhwc = is_nhwc(x);
   auto xc = nhwc ? x : x.contiguous();
   const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  // output dims: (H + 2*pad - k)/s (+ceil) + 1, pad=1, k=3
   auto odim = [&](int I) {
     const int num = I + 2 * 1 - 3;
     int o = (ceil_mode ? (num + (int)stride - 1) / (int)stride : num / (int)stride) + 1;
@@ -289,17 +430,20 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
   long long xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc;
   pool_strides(xc, nhwc, &xsb, &xsh, &xsw, &xsc);
   pool_strides(y, nhwc, &ysb, &ysh, &ysw, &ysc);
-  const long long total = (long long)B * OH * OW * C;
+  const int V = (nhwc && C % 4 == 0) ? 4 : 1;
+  const long long total = (long long)B * OH * OW * (C / V);
   auto stream = at::hip::getCurrentHIPStream();
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
-      x.scalar_type(), "maxpool3_fwd", [&] {
-    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
-    TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
-                "maxpool3: bf16/fp32 only");
-    maxpool3_fwd_kernel<T><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
-        idx.data_ptr<unsigned char>(), B, C, H, W, OH, OW, (int)stride,
-        xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
+  VISION_DISPATCH(x, "maxpool3_fwd", {
+    if (V == 4)
+      maxpool3_fwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
+          idx.data_ptr<unsigned char>(), B, C / 4, H, W, OH, OW, (int)stride,
+          xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
+    else
+      maxpool3_fwd_kernel<T, 1><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
+          idx.data_ptr<unsigned char>(), B, C, H, W, OH, OW, (int)stride,
+          xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
   });
   HIP_CHECK_LAST();
   return {y, idx};
@@ -319,17 +463,20 @@ torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
   long long xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc;
   pool_strides(dx, nhwc, &xsb, &xsh, &xsw, &xsc);
   pool_strides(dyc, nhwc, &ysb, &ysh, &ysw, &ysc);
-  const long long total = (long long)B * H * W * C;
+  const int V = (nhwc && C % 4 == 0) ? 4 : 1;
+  const long long total = (long long)B * H * W * (C / V);
   auto stream = at::hip::getCurrentHIPStream();
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
-      dy.scalar_type(), "maxpool3_bwd", [&] {
-    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
-    TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
-                "maxpool3: bf16/fp32 only");
-    maxpool3_bwd_kernel<T><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
-        reinterpret_cast<T*>(dx.data_ptr()), B, C, (int)H, (int)W, OH, OW, (int)stride,
-        xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
+  VISION_DISPATCH(dy, "maxpool3_bwd", {
+    if (V == 4)
+      maxpool3_bwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
+          reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW, (int)stride,
+          xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
+    else
+      maxpool3_bwd_kernel<T, 1><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
+          reinterpret_cast<T*>(dx.data_ptr()), B, C, (int)H, (int)W, OH, OW, (int)stride,
+          xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
   });
   HIP_CHECK_LAST();
   return dx;
