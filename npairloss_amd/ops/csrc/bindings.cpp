@@ -22,7 +22,7 @@ torch::Tensor bwd_weights(torch::Tensor S, torch::Tensor lab_l,
                           int64_t an_method, double scale);
 torch::Tensor recall_hits(torch::Tensor S, torch::Tensor lab_l,
                           torch::Tensor lab_g, int64_t rank,
-                          std::vector<int64_t> ks);
+                          torch::Tensor ks_t, int64_t kmax);
 
 // sort_select.hip
 torch::Tensor local_relative_thr(torch::Tensor S, torch::Tensor lab_l,

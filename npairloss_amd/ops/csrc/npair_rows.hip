@@ -334,24 +334,18 @@ torch::Tensor bwd_weights(torch::Tensor S, torch::Tensor lab_l,
 
 torch::Tensor recall_hits(torch::Tensor S, torch::Tensor lab_l,
                           torch::Tensor lab_g, int64_t rank,
-                          std::vector<int64_t> ks) {
+                          torch::Tensor ks_t, int64_t kmax) {
   check_sg(S, lab_l, lab_g);
   const int B = S.size(0), G = S.size(1);
-  const int nk = ks.size();
+  TORCH_CHECK(ks_t.is_cuda() && ks_t.dtype() == torch::kInt32 && ks_t.is_contiguous());
+  const int nk = ks_t.numel();
   TORCH_CHECK(nk >= 1 && nk <= 8, "1..8 k values");
-  int kmax = 0;
-  std::vector<int> ks32(nk);
-  for (int t = 0; t < nk; ++t) {
-    ks32[t] = (int)ks[t];
-    TORCH_CHECK(ks32[t] >= 1 && ks32[t] <= RECALL_MAX_TOPK, "k in [1,15]");
-    kmax = max(kmax, ks32[t]);
-  }
-  auto ks_t = torch::from_blob(ks32.data(), {nk}, torch::kInt32).to(S.device(), /*non_blocking=*/false);
+  TORCH_CHECK(kmax >= 1 && kmax <= RECALL_MAX_TOPK, "k in [1,15]");
   auto hits = torch::zeros({nk}, S.options().dtype(torch::kInt32));
   auto stream = at::hip::getCurrentHIPStream();
   recall_kernel<<<B, NPAIR_BLOCK, 0, stream>>>(
       S.data_ptr<float>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
-      (int)rank, ks_t.data_ptr<int>(), nk, kmax, hits.data_ptr<int>());
+      (int)rank, ks_t.data_ptr<int>(), nk, (int)kmax, hits.data_ptr<int>());
   HIP_CHECK_LAST();
   return hits;
 }

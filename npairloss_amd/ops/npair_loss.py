@@ -220,6 +220,18 @@ def _bwd_weights_torch(S, labels_l, labels_g, rank, thr_p, thr_n, max_all,
 # GPU (HIP extension) implementation
 # ---------------------------------------------------------------------------
 
+_KS_CACHE = {}
+
+
+def _ks_tensor(ks: tuple, device) -> torch.Tensor:
+    key = (ks, str(device))
+    t = _KS_CACHE.get(key)
+    if t is None:
+        t = torch.tensor(list(ks), dtype=torch.int32, device=device)
+        _KS_CACHE[key] = t
+    return t
+
+
 def _forward_hip(F_l, labels_l, F_g, labels_g, rank, cfg: NPairLossConfig, ks,
                  sim_dtype: str = "fp32"):
     C = _backend.ext()
@@ -262,7 +274,8 @@ def _forward_hip(F_l, labels_l, F_g, labels_g, rank, cfg: NPairLossConfig, ks,
         cfg.margin_ident, cfg.margin_diff,
         int(cfg.ap_mining_method), int(cfg.an_mining_method))
     loss = -log_term.sum() / B
-    recalls = C.recall_hits(S, lab_l, lab_g, rank, list(ks)).to(torch.float32) / B
+    ks_t = _ks_tensor(tuple(ks), S.device)
+    recalls = C.recall_hits(S, lab_l, lab_g, rank, ks_t, max(ks)).to(torch.float32) / B
     saved = dict(S=S, thr_p=thr_p.contiguous(), thr_n=thr_n.contiguous(), max_all=max_all_finite,
                  loss_ident=loss_ident, loss_sum=loss_sum,
                  ident_num=ident_num, diff_num=diff_num)

@@ -166,7 +166,8 @@ def test_recall_vs_torch(bg):
     B, G = bg
     S, lab_l, lab_g = rand_sg(B, G, ncls=6, seed=B + 1)
     ks = [1, 5, 10]
-    hits = _C().recall_hits(S, lab_l.int(), lab_g.int(), 0, ks)
+    ks_t = torch.tensor(ks, dtype=torch.int32, device="cuda")
+    hits = _C().recall_hits(S, lab_l.int(), lab_g.int(), 0, ks_t, max(ks))
     ref = NL._recall_torch(S, lab_l, lab_g, 0, ks) * B
     torch.testing.assert_close(hits.float(), ref, atol=0.01, rtol=0)
 
