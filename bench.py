@@ -38,6 +38,11 @@ def parse_args():
     p.add_argument("--ids-per-batch", type=int, default=None,
                    help="P identities per batch (default batch//2)")
     p.add_argument("--no-amp", action="store_true")
+    p.add_argument("--mining", choices=["production", "hard", "all"], default="production",
+                   help="production = GLOBAL RELATIVE_HARD ap + LOCAL HARD an (def.prototxt); "
+                        "hard = semi-hard negatives only; all = no mining (RAND)")
+    p.add_argument("--sim-dtype", choices=["fp32", "bf16", "fp8"], default="fp32",
+                   help="similarity-GEMM MFMA precision on GPU")
     p.add_argument("--timers", action="store_true",
                    help="per-phase HIP-event timing report on stderr (rank 0)")
     return p.parse_args()
@@ -77,11 +82,18 @@ def main():
     model = build_embedding_model(args.model).to(device)
     model = model.to(memory_format=torch.channels_last)
     model.train()
-    cfg = NPairLossConfig(
-        margin_ident=0.0, margin_diff=-0.05, identsn=-0.0, diffsn=-0.3,
-        ap_mining_region="GLOBAL", ap_mining_method="RELATIVE_HARD",
-        an_mining_region="LOCAL", an_mining_method="HARD")
-    loss_mod = NPairMultiClassLoss(cfg)
+    if args.mining == "production":
+        cfg = NPairLossConfig(
+            margin_ident=0.0, margin_diff=-0.05, identsn=-0.0, diffsn=-0.3,
+            ap_mining_region="GLOBAL", ap_mining_method="RELATIVE_HARD",
+            an_mining_region="LOCAL", an_mining_method="HARD")
+    elif args.mining == "hard":
+        cfg = NPairLossConfig(
+            margin_diff=-0.05, ap_mining_method="RAND",
+            an_mining_region="LOCAL", an_mining_method="HARD")
+    else:
+        cfg = NPairLossConfig()  # RAND/RAND: every pair
+    loss_mod = NPairMultiClassLoss(cfg, sim_dtype=args.sim_dtype)
     opt = CaffeSGD(model.parameters(), lr=0.001, momentum=0.9, weight_decay=2e-5)
     reducer = BucketedGradReducer(model)
     reducer.broadcast_params()
@@ -173,7 +185,8 @@ def main():
                 "batch_per_gpu": B,
                 "image": args.image,
                 "embed_dim": 1024 if args.model == "googlenet" else None,
-                "mining": "ap GLOBAL RELATIVE_HARD + an LOCAL HARD",
+                "mining": args.mining,
+                "sim_dtype": args.sim_dtype,
                 "parallelism": f"dp{world}",
                 "recall_top1_last_step": float(last_out["top1"]) if last_out else None,
             },
