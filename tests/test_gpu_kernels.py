@@ -214,3 +214,15 @@ def test_module_gpu_production_config_batch120():
     out.loss.backward()
     assert out.loss.item() == pytest.approx(fwds[0].loss, rel=2e-4, abs=1e-6)
     np.testing.assert_allclose(ft.grad.cpu().numpy(), grads[0], rtol=1e-3, atol=1e-6)
+
+
+def test_offline_recall_gpu_matches_cpu():
+    from npairloss_amd.eval import recall_at_k
+
+    f, lab = make_batch(num_classes=12, per_class=6, dim=64, seed=17)
+    ft = torch.from_numpy(f).float()
+    lt = torch.from_numpy(lab)
+    cpu = recall_at_k(ft, lt, ks=(1, 5, 10), chunk=24)
+    gpu = recall_at_k(ft.cuda(), lt.cuda(), ks=(1, 5, 10), chunk=24)
+    for k in (1, 5, 10):
+        assert gpu[k] == pytest.approx(cpu[k], abs=1e-9)
