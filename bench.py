@@ -45,6 +45,9 @@ def parse_args():
                    help="similarity-GEMM MFMA precision on GPU")
     p.add_argument("--timers", action="store_true",
                    help="per-phase HIP-event timing report on stderr (rank 0)")
+    p.add_argument("--graph", action="store_true",
+                   help="capture the whole train step in a hipGraph and replay "
+                        "(single GPU; removes per-kernel launch gaps)")
     return p.parse_args()
 
 
@@ -141,13 +144,43 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    graph = None
+    if args.graph:
+        if world > 1 or not use_cuda:
+            if rank == 0:
+                print("--graph requires single-GPU CUDA; ignoring", file=sys.stderr)
+        else:
+            # hipGraph capture of the full step (fwd + loss + bwd + optimizer):
+            # launch-bound inner loop becomes one graph replay.  Inputs stay
+            # at fixed addresses (batches[0]); MIOpen find must have run
+            # during warmup (cudnn.benchmark caches per shape).
+            for i in range(max(args.warmup, 3)):
+                step(0)
+            torch.cuda.synchronize()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                step(0)  # warm the allocator on the capture stream
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            timers.enabled = False  # timing events cannot record during capture
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                step(0)
+
+    def run_step(i):
+        if graph is not None:
+            graph.replay()
+        else:
+            step(i)
+
     for i in range(args.warmup):
-        step(i)
+        run_step(i)
     timers.report(reset=True)  # drop warmup phases
     barrier_sync()
     t0 = time.time()
     for i in range(args.steps):
-        step(args.warmup + i)
+        run_step(args.warmup + i)
     barrier_sync()
     elapsed = time.time() - t0
 
@@ -188,6 +221,7 @@ def main():
                 "mining": args.mining,
                 "sim_dtype": args.sim_dtype,
                 "parallelism": f"dp{world}",
+                "hip_graph": graph is not None,
                 "recall_top1_last_step": float(last_out["top1"]) if last_out else None,
             },
         }
