@@ -38,6 +38,9 @@ def parse_args():
     p.add_argument("--ids-per-batch", type=int, default=None,
                    help="P identities per batch (default batch//2)")
     p.add_argument("--no-amp", action="store_true")
+    p.add_argument("--pure-bf16", action="store_true",
+                   help="cast the model to bf16 with fp32 master weights in the "
+                        "optimizer (no per-step autocast weight casts)")
     p.add_argument("--mining", choices=["production", "hard", "all"], default="production",
                    help="production = GLOBAL RELATIVE_HARD ap + LOCAL HARD an (def.prototxt); "
                         "hard = semi-hard negatives only; all = no mining (RAND)")
@@ -84,6 +87,8 @@ def main():
 
     model = build_embedding_model(args.model).to(device)
     model = model.to(memory_format=torch.channels_last)
+    if args.pure_bf16:
+        model = model.to(torch.bfloat16)
     model.train()
     if args.mining == "production":
         cfg = NPairLossConfig(
@@ -97,11 +102,12 @@ def main():
     else:
         cfg = NPairLossConfig()  # RAND/RAND: every pair
     loss_mod = NPairMultiClassLoss(cfg, sim_dtype=args.sim_dtype)
-    opt = CaffeSGD(model.parameters(), lr=0.001, momentum=0.9, weight_decay=2e-5)
+    opt = CaffeSGD(model.parameters(), lr=0.001, momentum=0.9, weight_decay=2e-5,
+                   master_weights=args.pure_bf16)
     reducer = BucketedGradReducer(model)
     reducer.broadcast_params()
 
-    amp = use_cuda and not args.no_amp
+    amp = use_cuda and not args.no_amp and not args.pure_bf16
 
     # synthetic data: two alternating P x K-labelled image batches on device
     def make_batch(seed):
@@ -120,6 +126,8 @@ def main():
     def step(i):
         nonlocal last_out
         x, lab = batches[i % 2]
+        if args.pure_bf16:
+            x = x.to(torch.bfloat16)
         opt.zero_grad(set_to_none=True)
         with timers.phase("forward"):
             if amp:
@@ -210,7 +218,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if amp else "fp32",
+            "dtype": "bf16" if (amp or args.pure_bf16) else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,

@@ -19,9 +19,16 @@ from ..config.params import SolverConfig
 
 
 class CaffeSGD(torch.optim.Optimizer):
+    """Exact Caffe SGD.  With `master_weights=True` (for pure-bf16 models)
+    the optimizer keeps an fp32 master copy per parameter: gradients are
+    upcast, the velocity/update run in fp32, and the bf16 parameter is
+    refreshed from the master each step — removing autocast's ~2 weight
+    casts per conv per forward."""
+
     def __init__(self, params: Iterable, lr: float = 0.01, momentum: float = 0.0,
-                 weight_decay: float = 0.0):
-        defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay)
+                 weight_decay: float = 0.0, master_weights: bool = False):
+        defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay,
+                        master_weights=master_weights)
         super().__init__(params, defaults)
 
     @torch.no_grad()
@@ -37,21 +44,31 @@ class CaffeSGD(torch.optim.Optimizer):
             params = [p for p in group["params"] if p.grad is not None]
             if not params:
                 continue
+            master = group["master_weights"]
             grads = [p.grad for p in params]
-            vs = []
+            vs, ws = [], []
             for p in params:
                 st = self.state[p]
                 if "v" not in st:
-                    st["v"] = torch.zeros_like(p)
+                    if master and p.dtype != torch.float32:
+                        st["master"] = p.detach().float().clone()
+                    st["v"] = torch.zeros_like(st.get("master", p))
                 vs.append(st["v"])
+                ws.append(st.get("master", p))
+            if master:
+                grads = [g.float() if g.dtype != torch.float32 else g for g in grads]
             # fused multi-tensor update: v = mom*v + lr*g + (lr*wd)*w; w -= v
             # — 4 foreach kernels total, no temporaries (the reference-rule
             # lr*(g + wd*w) distributes so wd folds into a second add)
             torch._foreach_mul_(vs, mom)
             torch._foreach_add_(vs, grads, alpha=lr)
             if wd != 0:
-                torch._foreach_add_(vs, params, alpha=lr * wd)
-            torch._foreach_sub_(params, vs)
+                torch._foreach_add_(vs, ws, alpha=lr * wd)
+            torch._foreach_sub_(ws, vs)
+            if master:
+                for p, w in zip(params, ws):
+                    if w is not p:
+                        p.copy_(w)  # bf16 <- fp32 master
         return loss
 
     def set_lr(self, lr: float):

@@ -23,7 +23,8 @@ class _LRNFn(torch.autograd.Function):
             ctx.save_for_backward(x)
             return y
         ctx.save_for_backward(x)
-        return F.local_response_norm(x, size, alpha=alpha, beta=beta, k=k)
+        # CPU avg_pool3d lacks a bf16 kernel: compute in fp32, cast back
+        return F.local_response_norm(x.float(), size, alpha=alpha, beta=beta, k=k).to(x.dtype)
 
     @staticmethod
     def backward(ctx, dy):
@@ -33,9 +34,10 @@ class _LRNFn(torch.autograd.Function):
             dx = _backend.ext().lrn_bwd(x, dy, size, alpha, beta, k)
         else:
             with torch.enable_grad():
-                xr = x.detach().requires_grad_(True)
+                xr = x.detach().float().requires_grad_(True)
                 y = F.local_response_norm(xr, size, alpha=alpha, beta=beta, k=k)
-                (dx,) = torch.autograd.grad(y, xr, dy)
+                (dx,) = torch.autograd.grad(y, xr, dy.float())
+                dx = dx.to(x.dtype)
         return dx, None, None, None, None
 
 

@@ -245,3 +245,22 @@ def test_extract_embeddings():
     f, l = extract_embeddings(net, loader, device=torch.device("cpu"))
     assert f.shape == (32, 16) and l.shape == (32,)
     torch.testing.assert_close(f.norm(dim=1), torch.ones(32))
+
+
+def test_caffe_sgd_master_weights():
+    torch.manual_seed(0)
+    p32 = torch.nn.Parameter(torch.randn(64))
+    p16 = torch.nn.Parameter(p32.detach().bfloat16().clone())
+    o32 = CaffeSGD([p32], lr=0.1, momentum=0.9, weight_decay=0.01)
+    o16 = CaffeSGD([p16], lr=0.1, momentum=0.9, weight_decay=0.01, master_weights=True)
+    for i in range(5):
+        g = torch.randn(64)
+        p32.grad = g.clone()
+        p16.grad = g.bfloat16()
+        o32.step()
+        o16.step()
+    # fp32 master keeps the bf16 run close to the fp32 run (no drift beyond
+    # bf16 grad rounding)
+    torch.testing.assert_close(p16.float(), p32, rtol=5e-2, atol=5e-3)
+    master = o16.state[p16]["master"]
+    torch.testing.assert_close(p16.float(), master.to(torch.bfloat16).float())
