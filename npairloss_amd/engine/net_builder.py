@@ -70,6 +70,8 @@ def build_trainer_from_prototxt(
     amp_dtype: Optional[torch.dtype] = None,
     caffemodel: Optional[str] = None,
     num_workers: int = 0,
+    channels_last: Optional[bool] = None,
+    pure_bf16: bool = False,
 ) -> Trainer:
     net = parse_net_prototxt(_clean_prototxt(net_text))
 
@@ -140,7 +142,10 @@ def build_trainer_from_prototxt(
                                   test_spec.imgs_per_identity, seed=1)
         test_loader = torch.utils.data.DataLoader(tds, batch_sampler=tsampler, num_workers=num_workers)
 
+    if channels_last is None:
+        channels_last = torch.cuda.is_available()
     trainer = Trainer(model, loss_mod, solver, loader, test_loader=test_loader,
-                      device=device, amp_dtype=amp_dtype)
+                      device=device, amp_dtype=amp_dtype,
+                      channels_last=channels_last, pure_bf16=pure_bf16)
     trainer.augment = aug  # applied by train CLI before the model (TRAIN phase)
     return trainer

@@ -76,6 +76,8 @@ class CaffeSGD(torch.optim.Optimizer):
             group["lr"] = lr
 
 
-def build_optimizer(model: torch.nn.Module, solver: SolverConfig) -> CaffeSGD:
+def build_optimizer(model: torch.nn.Module, solver: SolverConfig,
+                    master_weights: bool = False) -> CaffeSGD:
     return CaffeSGD(model.parameters(), lr=solver.base_lr,
-                    momentum=solver.momentum, weight_decay=solver.weight_decay)
+                    momentum=solver.momentum, weight_decay=solver.weight_decay,
+                    master_weights=master_weights)

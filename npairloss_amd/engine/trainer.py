@@ -45,20 +45,24 @@ class Trainer:
                  device: Optional[torch.device] = None,
                  amp_dtype: Optional[torch.dtype] = None,
                  log_fn: Callable[[str], None] = print,
-                 channels_last: bool = False):
+                 channels_last: bool = False,
+                 pure_bf16: bool = False):
         self.device = device or (torch.device("cuda") if torch.cuda.is_available()
                                  else torch.device("cpu"))
         self.model = model.to(self.device)
         if channels_last:
             self.model = self.model.to(memory_format=torch.channels_last)
         self.channels_last = channels_last
+        self.pure_bf16 = pure_bf16
+        if pure_bf16:
+            self.model = self.model.to(torch.bfloat16)
         self.loss = loss
         self.solver = solver
         self.train_loader = train_loader
         self.test_loader = test_loader
         self.amp_dtype = amp_dtype
         self.log = log_fn
-        self.optimizer = build_optimizer(self.model, solver)
+        self.optimizer = build_optimizer(self.model, solver, master_weights=pure_bf16)
         self.reducer = BucketedGradReducer(self.model)
         self.reducer.broadcast_params()
         self.iter = 0
@@ -98,6 +102,8 @@ class Trainer:
         if self.channels_last and images.dim() == 4:
             images = images.to(memory_format=torch.channels_last)
         labels = labels.to(self.device, non_blocking=True)
+        if self.pure_bf16:
+            images = images.to(torch.bfloat16)
         lr = self.solver.lr_at(self.iter)
         self.optimizer.set_lr(lr)
         self.optimizer.zero_grad(set_to_none=True)

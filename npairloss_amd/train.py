@@ -24,6 +24,8 @@ def main(argv=None):
     p.add_argument("--max-iter", type=int, default=None)
     p.add_argument("--synthetic-classes", type=int, default=256)
     p.add_argument("--amp", choices=["off", "bf16", "fp16"], default="bf16")
+    p.add_argument("--pure-bf16", action="store_true",
+                   help="bf16 model + fp32 master weights (replaces --amp)")
     p.add_argument("--num-workers", type=int, default=2)
     args = p.parse_args(argv)
 
@@ -46,12 +48,13 @@ def main(argv=None):
         if torch.cuda.is_available():
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
 
-    amp_dtype = {"off": None, "bf16": torch.bfloat16, "fp16": torch.float16}[args.amp]
+    amp_dtype = None if args.pure_bf16 else {"off": None, "bf16": torch.bfloat16,
+                                              "fp16": torch.float16}[args.amp]
     caffemodel = args.weights if (args.weights and args.weights.endswith(".caffemodel")) else None
     trainer = build_trainer_from_prototxt(
         net_text, solver, synthetic_classes=args.synthetic_classes,
         amp_dtype=amp_dtype, caffemodel=caffemodel,
-        num_workers=args.num_workers)
+        num_workers=args.num_workers, pure_bf16=args.pure_bf16)
     if args.weights and args.weights.endswith(".pt"):
         trainer.restore(args.weights)
     trainer.fit(max_iter=args.max_iter)
