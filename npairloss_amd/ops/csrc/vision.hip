@@ -182,11 +182,6 @@ __global__ void lrn_bwd_generic(const T* __restrict__ x, const T* __restrict__ d
 // 3x3 max pool, pad 1, stride 1 or 2 — V channels per thread (V = 4 or 1)
 // ---------------------------------------------------------------------------
 
-// Thread assignment: grid-stride over the (OH*OW*Cv) SPATIAL sites; each
-// thread then loops over the batch at a fixed stride.  The index
-// decomposition (3 int64 div/mods) and the window-bounds logic run ONCE
-// per site instead of once per element — the inner b-loop is pure pointer
-// arithmetic + loads.
 template <typename T, int V>
 __global__ void maxpool3_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                     unsigned char* __restrict__ idx,
@@ -194,19 +189,26 @@ __global__ void maxpool3_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                     int stride,
                                     long long xsb, long long xsh, long long xsw, long long xsc,
                                     long long ysb, long long ysh, long long ysw, long long ysc) {
-  const long long sites = (long long)OH * OW * Cv;
-  for (long long s = (long long)blockIdx.x * blockDim.x + threadIdx.x; s < sites;
-       s += (long long)gridDim.x * blockDim.x) {
-    const int cv = s % Cv;
-    long long r = s / Cv;
+  // Cv = C / V; i enumerates ((b*OH + oh)*OW + ow)*Cv + cv  (cv fastest)
+  const long long total = (long long)B * OH * OW * Cv;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    const int cv = i % Cv;
+    long long r = i / Cv;
     const int ow = r % OW;
-    const int oh = r / OW;
+    r /= OW;
+    const int oh = r % OH;
+    const int b = r / OH;
     const int h0 = oh * stride - 1;
     const int w0 = ow * stride - 1;
-    // in-bounds window taps, precomputed as x-offsets relative to (b, cv)
-    long long taps[9];
-    int tapid[9];
-    int ntap = 0;
+    float best[V];
+    int besti[V];
+#pragma unroll
+    for (int v = 0; v < V; ++v) {
+      best[v] = -FLT_MAX;
+      besti[v] = 0;
+    }
+    const long long xb = (long long)b * xsb + (long long)(cv * V) * xsc;
 #pragma unroll
     for (int dh = 0; dh < 3; ++dh) {
       const int h = h0 + dh;
@@ -215,48 +217,31 @@ __global__ void maxpool3_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
       for (int dw = 0; dw < 3; ++dw) {
         const int w = w0 + dw;
         if (w < 0 || w >= W) continue;
-        taps[ntap] = (long long)h * xsh + (long long)w * xsw;
-        tapid[ntap] = dh * 3 + dw;
-        ++ntap;
-      }
-    }
-    const int bper = (B + gridDim.y - 1) / gridDim.y;
-    const int b0 = blockIdx.y * bper;
-    const int b1 = min(B, b0 + bper);
-    long long xb = (long long)(cv * V) * xsc + (long long)b0 * xsb;
-    long long yi = (long long)oh * ysh + (long long)ow * ysw + (long long)(cv * V) * ysc
-                 + (long long)b0 * ysb;
-    for (int b = b0; b < b1; ++b, xb += xsb, yi += ysb) {
-      float best[V];
-      int besti[V];
-#pragma unroll
-      for (int v = 0; v < V; ++v) {
-        best[v] = -FLT_MAX;
-        besti[v] = 0;
-      }
-      for (int t = 0; t < ntap; ++t) {
+        const long long a = xb + (long long)h * xsh + (long long)w * xsw;
         float vals[V];
         if (V == 4) {
-          ld4(x + xb + taps[t], vals);
+          ld4(x + a, vals);
         } else {
-          vals[0] = ldf(x, xb + taps[t]);
+          vals[0] = ldf(x, a);
         }
 #pragma unroll
         for (int v = 0; v < V; ++v)
           if (vals[v] > best[v]) {
             best[v] = vals[v];
-            besti[v] = tapid[t];
+            besti[v] = dh * 3 + dw;
           }
       }
-      if (V == 4) {
-        st4(y + yi, best);
-        uchar4 u;
-        u.x = besti[0]; u.y = besti[1]; u.z = besti[2]; u.w = besti[3];
-        *reinterpret_cast<uchar4*>(idx + yi) = u;
-      } else {
-        stf(y, yi, best[0]);
-        idx[yi] = (unsigned char)besti[0];
-      }
+    }
+    const long long yi = (long long)b * ysb + (long long)oh * ysh + (long long)ow * ysw
+                       + (long long)(cv * V) * ysc;
+    if (V == 4) {
+      st4(y + yi, best);
+      uchar4 u;
+      u.x = besti[0]; u.y = besti[1]; u.z = besti[2]; u.w = besti[3];
+      *reinterpret_cast<uchar4*>(idx + yi) = u;
+    } else {
+      stf(y, yi, best[0]);
+      idx[yi] = (unsigned char)besti[0];
     }
   }
 }
@@ -269,67 +254,54 @@ __global__ void maxpool3_bwd_kernel(const T* __restrict__ dy,
                                     int stride,
                                     long long xsb, long long xsh, long long xsw, long long xsc,
                                     long long ysb, long long ysh, long long ysw, long long ysc) {
-  // same site-outer / batch-inner structure as the forward: the covering
-  // windows of (h, w) and their tap codes are computed once per site
-  const long long sites = (long long)H * W * Cv;
-  for (long long s = (long long)blockIdx.x * blockDim.x + threadIdx.x; s < sites;
-       s += (long long)gridDim.x * blockDim.x) {
-    const int cv = s % Cv;
-    long long r = s / Cv;
+  const long long total = (long long)B * H * W * Cv;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long long)gridDim.x * blockDim.x) {
+    const int cv = i % Cv;
+    long long r = i / Cv;
     const int w = r % W;
-    const int h = r / W;
+    r /= W;
+    const int h = r % H;
+    const int b = r / H;
+    float acc[V];
+#pragma unroll
+    for (int v = 0; v < V; ++v) acc[v] = 0.f;
+    const long long yb = (long long)b * ysb + (long long)(cv * V) * ysc;
     // output windows covering (h, w): oh*stride - 1 <= h <= oh*stride + 1,
     // i.e. oh in [ceil((h-1)/s), floor((h+1)/s)] (negative lower clamps to 0)
     const int oh_lo = max(0, (h - 1 + stride - 1) / stride);
     const int oh_hi = min(OH - 1, (h + 1) / stride);
     const int ow_lo = max(0, (w - 1 + stride - 1) / stride);
     const int ow_hi = min(OW - 1, (w + 1) / stride);
-    long long taps[9];
-    unsigned char want[9];
-    int ntap = 0;
     for (int oh = oh_lo; oh <= oh_hi; ++oh) {
       const int dh = h - (oh * stride - 1);
       if (dh < 0 || dh > 2) continue;
       for (int ow = ow_lo; ow <= ow_hi; ++ow) {
         const int dw = w - (ow * stride - 1);
         if (dw < 0 || dw > 2) continue;
-        taps[ntap] = (long long)oh * ysh + (long long)ow * ysw;
-        want[ntap] = (unsigned char)(dh * 3 + dw);
-        ++ntap;
-      }
-    }
-    const int bper = (B + gridDim.y - 1) / gridDim.y;
-    const int b0 = blockIdx.y * bper;
-    const int b1 = min(B, b0 + bper);
-    long long yb = (long long)(cv * V) * ysc + (long long)b0 * ysb;
-    long long xi = (long long)h * xsh + (long long)w * xsw + (long long)(cv * V) * xsc
-                 + (long long)b0 * xsb;
-    for (int b = b0; b < b1; ++b, yb += ysb, xi += xsb) {
-      float acc[V];
-#pragma unroll
-      for (int v = 0; v < V; ++v) acc[v] = 0.f;
-      for (int t = 0; t < ntap; ++t) {
-        const long long yi = yb + taps[t];
-        const unsigned char wa = want[t];
+        const long long yi = yb + (long long)oh * ysh + (long long)ow * ysw;
+        const unsigned char want = (unsigned char)(dh * 3 + dw);
         if (V == 4) {
           const uchar4 u = *reinterpret_cast<const uchar4*>(idx + yi);
-          if (u.x == wa || u.y == wa || u.z == wa || u.w == wa) {
+          if (u.x == want || u.y == want || u.z == want || u.w == want) {
             float g[4];
             ld4(dy + yi, g);
-            if (u.x == wa) acc[0] += g[0];
-            if (u.y == wa) acc[1] += g[1];
-            if (u.z == wa) acc[2] += g[2];
-            if (u.w == wa) acc[3] += g[3];
+            if (u.x == want) acc[0] += g[0];
+            if (u.y == want) acc[1] += g[1];
+            if (u.z == want) acc[2] += g[2];
+            if (u.w == want) acc[3] += g[3];
           }
         } else {
-          if (idx[yi] == wa) acc[0] += ldf(dy, yi);
+          if (idx[yi] == want) acc[0] += ldf(dy, yi);
         }
       }
-      if (V == 4) {
-        st4(dx + xi, acc);
-      } else {
-        stf(dx, xi, acc[0]);
-      }
+    }
+    const long long xi = (long long)b * xsb + (long long)h * xsh + (long long)w * xsw
+                       + (long long)(cv * V) * xsc;
+    if (V == 4) {
+      st4(dx + xi, acc);
+    } else {
+      stf(dx, xi, acc[0]);
     }
   }
 }
@@ -459,18 +431,16 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
   pool_strides(xc, nhwc, &xsb, &xsh, &xsw, &xsc);
   pool_strides(y, nhwc, &ysb, &ysh, &ysw, &ysc);
   const int V = (nhwc && C % 4 == 0) ? 4 : 1;
-  const long long sites = (long long)OH * OW * (C / V);
-  const int nby = (int)std::min<long long>(B, std::max<long long>(1, (512LL * 1024) / std::max<long long>(sites, 1)));
-  dim3 grid(grid_for(sites), nby);
+  const long long total = (long long)B * OH * OW * (C / V);
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(x, "maxpool3_fwd", {
     if (V == 4)
-      maxpool3_fwd_kernel<T, 4><<<grid, NPAIR_BLOCK, 0, stream>>>(
+      maxpool3_fwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
           idx.data_ptr<unsigned char>(), B, C / 4, H, W, OH, OW, (int)stride,
           xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
     else
-      maxpool3_fwd_kernel<T, 1><<<grid, NPAIR_BLOCK, 0, stream>>>(
+      maxpool3_fwd_kernel<T, 1><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
           idx.data_ptr<unsigned char>(), B, C, H, W, OH, OW, (int)stride,
           xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
@@ -494,18 +464,16 @@ torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
   pool_strides(dx, nhwc, &xsb, &xsh, &xsw, &xsc);
   pool_strides(dyc, nhwc, &ysb, &ysh, &ysw, &ysc);
   const int V = (nhwc && C % 4 == 0) ? 4 : 1;
-  const long long sites = (long long)H * W * (C / V);
-  const int nby = (int)std::min<long long>(B, std::max<long long>(1, (512LL * 1024) / std::max<long long>(sites, 1)));
-  dim3 grid(grid_for(sites), nby);
+  const long long total = (long long)B * H * W * (C / V);
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(dy, "maxpool3_bwd", {
     if (V == 4)
-      maxpool3_bwd_kernel<T, 4><<<grid, NPAIR_BLOCK, 0, stream>>>(
+      maxpool3_bwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
           reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW, (int)stride,
           xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
     else
-      maxpool3_bwd_kernel<T, 1><<<grid, NPAIR_BLOCK, 0, stream>>>(
+      maxpool3_bwd_kernel<T, 1><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
           reinterpret_cast<T*>(dx.data_ptr()), B, C, (int)H, (int)W, OH, OW, (int)stride,
           xsb, xsh, xsw, xsc, ysb, ysh, ysw, ysc);
