@@ -17,7 +17,7 @@ from typing import Optional
 
 import torch
 
-from ..config.params import (NetConfig, NPairLossConfig, SolverConfig,
+from ..config.params import (NPairLossConfig, SolverConfig,
                              parse_net_prototxt)
 from ..data import PKBatchSampler, SyntheticImageDataset
 from ..data.transforms import DataTransformer, TransformConfig
@@ -102,10 +102,10 @@ def build_trainer_from_prototxt(
         aug_cfg = TransformConfig.from_message(dt_layers[0].raw.get("data_transformer_l_param"))
         aug = DataTransformer(aug_cfg)
 
-    # --- backbone: the reference's conv stack is GoogLeNet v1
-    conv_names = [l.name for l in net.layers if l.type == "Convolution"]
+    # --- backbone: the reference's conv stack is GoogLeNet v1;
+    # L2-normalize the embedding iff the net has an L2Normalize layer
     has_l2 = bool(net.find("L2Normalize"))
-    model = build_embedding_model("googlenet", normalize=has_l2 or True)
+    model = build_embedding_model("googlenet", normalize=has_l2)
     if caffemodel:
         load_caffemodel_into(model, caffemodel)
 

@@ -24,7 +24,7 @@ from ..config.params import SolverConfig
 from ..ops.npair_loss import NPairMultiClassLoss
 from ..parallel import collectives as comm
 from ..parallel.ddp import BucketedGradReducer
-from .solver import CaffeSGD, build_optimizer
+from .solver import build_optimizer
 
 
 class AverageWindow:

@@ -29,7 +29,6 @@ database entries, self excluded):
 
 from __future__ import annotations
 
-import math
 from typing import NamedTuple, Optional, Sequence, Tuple
 
 import torch
