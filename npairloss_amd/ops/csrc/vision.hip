@@ -126,22 +126,34 @@ __global__ void lrn_tile_kernel(const T* __restrict__ x, const T* __restrict__ d
   }
 }
 
-// 4-channels-per-thread variant (NHWC, C % 4 == 0, C <= 1024): 8-byte bf16
-// loads/stores and a quarter of the LDS instructions of the scalar tile
-// kernel; still one powf per element.
+// 4-channels-per-thread variant (NHWC, C % 4 == 0, n <= 9, C <= 1024):
+// 8-byte bf16 global loads/stores; the channel rows live in LDS with a
+// 4-float zero halo on each side so each thread's window sums come from
+// THREE aligned float4 reads (conflict-free ds_read_b128 — the naive
+// per-channel scalar reads at stride 4 are a 4-way bank conflict).
+// Out-of-range window taps read halo zeros == the clamp semantics.
+#define LRN_HALO 4
+
 template <typename T, bool BWD>
 __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                  T* __restrict__ out, long long npix, int C,
                                  int n, float alpha_over_n, float beta, float k) {
-  __shared__ float xsq[NPAIR_BLOCK * 4];
-  __shared__ float tbuf[NPAIR_BLOCK * 4];
+  __shared__ float xsq[NPAIR_BLOCK * 5];
+  __shared__ float tbuf[NPAIR_BLOCK * 5];
   const int half = n / 2;
   const int C_q = C / 4;                      // threads per pixel
   const int ppb = NPAIR_BLOCK / C_q;          // pixels per block
+  const int row = C + 2 * LRN_HALO;           // padded LDS row per pixel
   const int pl = threadIdx.x / C_q;
   const int c0 = (threadIdx.x % C_q) * 4;
-  const int lds0 = pl * C;
+  const int lds0 = pl * row + LRN_HALO;
   const bool chan_ok = pl < ppb;
+  // zero the halos (and everything) once; interiors are rewritten per tile
+  for (int i = threadIdx.x; i < ppb * row; i += blockDim.x) {
+    xsq[i] = 0.f;
+    tbuf[i] = 0.f;
+  }
+  __syncthreads();
   const long long tiles = (npix + ppb - 1) / ppb;
   for (long long tile = blockIdx.x; tile < tiles; tile += gridDim.x) {
     const long long pix = tile * ppb + pl;
@@ -153,19 +165,25 @@ __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ 
       ld4(x + base, xi);
       if (BWD) ld4(dy + base, gi);
     }
-    if (chan_ok) {
-#pragma unroll
-      for (int v = 0; v < 4; ++v) xsq[lds0 + c0 + v] = xi[v] * xi[v];
-    }
+    if (chan_ok)
+      *reinterpret_cast<float4*>(&xsq[lds0 + c0]) =
+          make_float4(xi[0] * xi[0], xi[1] * xi[1], xi[2] * xi[2], xi[3] * xi[3]);
     __syncthreads();
+    // window sums for channels [c0, c0+3] from the 12-float neighborhood
+    float buf[12];
+    {
+      const float4 a = *reinterpret_cast<const float4*>(&xsq[lds0 + c0 - 4]);
+      const float4 b = *reinterpret_cast<const float4*>(&xsq[lds0 + c0]);
+      const float4 c = *reinterpret_cast<const float4*>(&xsq[lds0 + c0 + 4]);
+      buf[0] = a.x; buf[1] = a.y; buf[2] = a.z; buf[3] = a.w;
+      buf[4] = b.x; buf[5] = b.y; buf[6] = b.z; buf[7] = b.w;
+      buf[8] = c.x; buf[9] = c.y; buf[10] = c.z; buf[11] = c.w;
+    }
     float scale[4], p[4];
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
-      const int c = c0 + v;
-      const int lo = max(0, c - half);
-      const int hi = min(C - 1, c + half);
       float ss = 0.f;
-      for (int j = lo; j <= hi; ++j) ss += xsq[lds0 + j];
+      for (int j = -half; j <= half; ++j) ss += buf[4 + v + j];  // halo zeros = clamp
       scale[v] = k + alpha_over_n * ss;
       p[v] = __powf(scale[v], -beta);
     }
@@ -179,21 +197,25 @@ __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ 
       __syncthreads();
       continue;
     }
-    if (chan_ok) {
-#pragma unroll
-      for (int v = 0; v < 4; ++v)
-        tbuf[lds0 + c0 + v] = act ? gi[v] * (xi[v] * p[v]) / scale[v] : 0.f;
-    }
+    if (chan_ok)
+      *reinterpret_cast<float4*>(&tbuf[lds0 + c0]) =
+          make_float4(gi[0] * (xi[0] * p[0]) / scale[0], gi[1] * (xi[1] * p[1]) / scale[1],
+                      gi[2] * (xi[2] * p[2]) / scale[2], gi[3] * (xi[3] * p[3]) / scale[3]);
     __syncthreads();
+    {
+      const float4 a = *reinterpret_cast<const float4*>(&tbuf[lds0 + c0 - 4]);
+      const float4 b = *reinterpret_cast<const float4*>(&tbuf[lds0 + c0]);
+      const float4 c = *reinterpret_cast<const float4*>(&tbuf[lds0 + c0 + 4]);
+      buf[0] = a.x; buf[1] = a.y; buf[2] = a.z; buf[3] = a.w;
+      buf[4] = b.x; buf[5] = b.y; buf[6] = b.z; buf[7] = b.w;
+      buf[8] = c.x; buf[9] = c.y; buf[10] = c.z; buf[11] = c.w;
+    }
     if (act) {
       float dxv[4];
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
-        const int c = c0 + v;
-        const int lo = max(0, c - half);
-        const int hi = min(C - 1, c + half);
         float cross = 0.f;
-        for (int j = lo; j <= hi; ++j) cross += tbuf[lds0 + j];
+        for (int j = -half; j <= half; ++j) cross += buf[4 + v + j];
         dxv[v] = gi[v] * p[v] - 2.f * alpha_over_n * beta * xi[v] * cross;
       }
       st4(out + base, dxv);
@@ -414,7 +436,7 @@ torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, 
   const long long cstride = nhwc ? 1 : S;
   auto stream = at::hip::getCurrentHIPStream();
   const float aon = (float)(alpha / size);
-  if (nhwc && C % 4 == 0 && C <= 1024) {
+  if (nhwc && C % 4 == 0 && C <= 1024 && size <= 9 && k > 0) {
     const long long npix = B * S;
     const int ppb = NPAIR_BLOCK / (int)(C / 4);
     const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
@@ -459,7 +481,7 @@ torch::Tensor lrn_bwd(torch::Tensor x, torch::Tensor dy, int64_t size, double al
   const long long cstride = nhwc ? 1 : S;
   auto stream = at::hip::getCurrentHIPStream();
   const float aon = (float)(alpha / size);
-  if (nhwc && C % 4 == 0 && C <= 1024) {
+  if (nhwc && C % 4 == 0 && C <= 1024 && size <= 9 && k > 0) {
     const long long npix = B * S;
     const int ppb = NPAIR_BLOCK / (int)(C / 4);
     const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
