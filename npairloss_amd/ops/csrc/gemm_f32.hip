@@ -33,8 +33,11 @@ __global__ void gemm_f32_kernel(const float* __restrict__ A,
                                 const float* __restrict__ B,
                                 float* __restrict__ C,
                                 int M, int N, int K, float alpha) {
-  __shared__ float As[GEMM_BK][GEMM_BM];
-  __shared__ float Bs[GEMM_BK][GEMM_BN];
+  // +1 row padding: unpadded 64-float rows put the 4 staging writes of
+  // each lane quartet on ONE bank (4-way conflict; PMC showed ~2M conflict
+  // cycles vs 0.5M MFMA instructions per dispatch)
+  __shared__ float As[GEMM_BK][GEMM_BM + 1];
+  __shared__ float Bs[GEMM_BK][GEMM_BN + 1];
   const int m0 = blockIdx.y * GEMM_BM;
   const int n0 = blockIdx.x * GEMM_BN;
   const int t = threadIdx.x;

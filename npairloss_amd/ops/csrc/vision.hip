@@ -134,13 +134,13 @@ __global__ void lrn_tile_kernel(const T* __restrict__ x, const T* __restrict__ d
 // Out-of-range window taps read halo zeros == the clamp semantics.
 #define LRN_HALO 4
 
-template <typename T, bool BWD>
+template <typename T, bool BWD, int HALF>
 __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ dy,
                                  T* __restrict__ out, long long npix, int C,
                                  int n, float alpha_over_n, float beta, float k) {
   __shared__ float xsq[NPAIR_BLOCK * 5];
   __shared__ float tbuf[NPAIR_BLOCK * 5];
-  const int half = n / 2;
+  constexpr int half = HALF;
   const int C_q = C / 4;                      // threads per pixel
   const int ppb = NPAIR_BLOCK / C_q;          // pixels per block
   const int row = C + 2 * LRN_HALO;           // padded LDS row per pixel
@@ -183,6 +183,7 @@ __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ 
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
       float ss = 0.f;
+#pragma unroll
       for (int j = -half; j <= half; ++j) ss += buf[4 + v + j];  // halo zeros = clamp
       scale[v] = k + alpha_over_n * ss;
       p[v] = __powf(scale[v], -beta);
@@ -215,6 +216,7 @@ __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ 
 #pragma unroll
       for (int v = 0; v < 4; ++v) {
         float cross = 0.f;
+#pragma unroll
         for (int j = -half; j <= half; ++j) cross += buf[4 + v + j];
         dxv[v] = gi[v] * p[v] - 2.f * alpha_over_n * beta * xi[v] * cross;
       }
@@ -287,16 +289,18 @@ __global__ void maxpool3_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                     int stride,
                                     long long xsb, long long xsh, long long xsw, long long xsc,
                                     long long ysb, long long ysh, long long ysw, long long ysc) {
-  // Cv = C / V; i enumerates ((b*OH + oh)*OW + ow)*Cv + cv  (cv fastest)
-  const long long total = (long long)B * OH * OW * Cv;
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long long)gridDim.x * blockDim.x) {
-    const int cv = i % Cv;
-    long long r = i / Cv;
-    const int ow = r % OW;
-    r /= OW;
-    const int oh = r % OH;
-    const int b = r / OH;
+  // Cv = C / V; i enumerates ((b*OH + oh)*OW + ow)*Cv + cv  (cv fastest).
+  // All supported shapes fit 32 bits -> uint division (the 64-bit div/mod
+  // chain showed up as ~40 VALU instructions per element in PMC counts).
+  const unsigned int total = (unsigned int)((long long)B * OH * OW * Cv);
+  for (unsigned int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    const int cv = i % (unsigned int)Cv;
+    unsigned int r = i / (unsigned int)Cv;
+    const int ow = r % (unsigned int)OW;
+    r /= (unsigned int)OW;
+    const int oh = r % (unsigned int)OH;
+    const int b = r / (unsigned int)OH;
     const int h0 = oh * stride - 1;
     const int w0 = ow * stride - 1;
     float best[V];
@@ -352,15 +356,15 @@ __global__ void maxpool3_bwd_kernel(const T* __restrict__ dy,
                                     int stride,
                                     long long xsb, long long xsh, long long xsw, long long xsc,
                                     long long ysb, long long ysh, long long ysw, long long ysc) {
-  const long long total = (long long)B * H * W * Cv;
-  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long long)gridDim.x * blockDim.x) {
-    const int cv = i % Cv;
-    long long r = i / Cv;
-    const int w = r % W;
-    r /= W;
-    const int h = r % H;
-    const int b = r / H;
+  const unsigned int total = (unsigned int)((long long)B * H * W * Cv);
+  for (unsigned int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    const int cv = i % (unsigned int)Cv;
+    unsigned int r = i / (unsigned int)Cv;
+    const int w = r % (unsigned int)W;
+    r /= (unsigned int)W;
+    const int h = r % (unsigned int)H;
+    const int b = r / (unsigned int)H;
     float acc[V];
 #pragma unroll
     for (int v = 0; v < V; ++v) acc[v] = 0.f;
@@ -440,11 +444,21 @@ torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, 
     const long long npix = B * S;
     const int ppb = NPAIR_BLOCK / (int)(C / 4);
     const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
+    const int half = (int)size / 2;
     VISION_DISPATCH(x, "lrn_fwd", {
-      lrn_tile4_kernel<T, false><<<blocks, NPAIR_BLOCK, 0, stream>>>(
-          reinterpret_cast<const T*>(xc.data_ptr()), nullptr,
-          reinterpret_cast<T*>(y.data_ptr()), npix, (int)C,
-          (int)size, aon, (float)beta, (float)k);
+      auto launch = [&](auto kfn) {
+        kfn<<<blocks, NPAIR_BLOCK, 0, stream>>>(
+            reinterpret_cast<const T*>(xc.data_ptr()), nullptr,
+            reinterpret_cast<T*>(y.data_ptr()), npix, (int)C,
+            (int)size, aon, (float)beta, (float)k);
+      };
+      switch (half) {
+        case 0: launch(lrn_tile4_kernel<T, false, 0>); break;
+        case 1: launch(lrn_tile4_kernel<T, false, 1>); break;
+        case 2: launch(lrn_tile4_kernel<T, false, 2>); break;
+        case 3: launch(lrn_tile4_kernel<T, false, 3>); break;
+        default: launch(lrn_tile4_kernel<T, false, 4>); break;
+      }
     });
   } else if (C <= NPAIR_BLOCK) {
     const int C_pad = (int)((C + 63) / 64) * 64;
@@ -485,12 +499,22 @@ torch::Tensor lrn_bwd(torch::Tensor x, torch::Tensor dy, int64_t size, double al
     const long long npix = B * S;
     const int ppb = NPAIR_BLOCK / (int)(C / 4);
     const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
+    const int half = (int)size / 2;
     VISION_DISPATCH(x, "lrn_bwd", {
-      lrn_tile4_kernel<T, true><<<blocks, NPAIR_BLOCK, 0, stream>>>(
-          reinterpret_cast<const T*>(xc.data_ptr()),
-          reinterpret_cast<const T*>(dyc.data_ptr()),
-          reinterpret_cast<T*>(dx.data_ptr()), npix, (int)C,
-          (int)size, aon, (float)beta, (float)k);
+      auto launch = [&](auto kfn) {
+        kfn<<<blocks, NPAIR_BLOCK, 0, stream>>>(
+            reinterpret_cast<const T*>(xc.data_ptr()),
+            reinterpret_cast<const T*>(dyc.data_ptr()),
+            reinterpret_cast<T*>(dx.data_ptr()), npix, (int)C,
+            (int)size, aon, (float)beta, (float)k);
+      };
+      switch (half) {
+        case 0: launch(lrn_tile4_kernel<T, true, 0>); break;
+        case 1: launch(lrn_tile4_kernel<T, true, 1>); break;
+        case 2: launch(lrn_tile4_kernel<T, true, 2>); break;
+        case 3: launch(lrn_tile4_kernel<T, true, 3>); break;
+        default: launch(lrn_tile4_kernel<T, true, 4>); break;
+      }
     });
   } else if (C <= NPAIR_BLOCK) {
     const int C_pad = (int)((C + 63) / 64) * 64;
