@@ -69,6 +69,9 @@ class Trainer:
         self.avg = AverageWindow(solver.average_loss)
         self.history = []
         self.augment = None  # optional TRAIN-phase batch augmentation (DataTransformer)
+        # failure detection: raise on NaN/Inf loss every N iters (0 = off).
+        # (the reference had none — an NaN would silently poison the run)
+        self.divergence_check = 25
 
     # -- checkpointing ------------------------------------------------------
 
@@ -118,6 +121,11 @@ class Trainer:
         self.reducer.finalize()
         self.optimizer.step()
         self.iter += 1
+        if self.divergence_check and self.iter % self.divergence_check == 0:
+            lv = float(out.loss)
+            if not (lv == lv and abs(lv) != float("inf")):
+                raise FloatingPointError(
+                    f"training diverged: loss={lv} at iter {self.iter}")
         return {
             "loss": out.loss, "top1": out.retrieve_top1, "top5": out.retrieve_top5,
             "top10": out.retrieve_top10, "asum": out.feature_asum, "lr": lr,

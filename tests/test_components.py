@@ -264,3 +264,38 @@ def test_caffe_sgd_master_weights():
     torch.testing.assert_close(p16.float(), p32, rtol=5e-2, atol=5e-3)
     master = o16.state[p16]["master"]
     torch.testing.assert_close(p16.float(), master.to(torch.bfloat16).float())
+
+
+def test_trainer_divergence_guard():
+    tr = _tiny_trainer()
+    tr.divergence_check = 1
+
+    class _BadLoss(torch.nn.Module):
+        def forward(self, f, l):
+            from npairloss_amd.ops.npair_loss import NPairLossOutput
+            nan = (f.sum() * float("nan"))
+            z = torch.zeros(())
+            return NPairLossOutput(nan, z, z, z, z)
+
+    tr.loss = _BadLoss()
+    with pytest.raises(FloatingPointError):
+        tr.fit(max_iter=1)
+
+
+def test_export_cli_roundtrip(tmp_path):
+    from npairloss_amd.export import main as export_main
+    from npairloss_amd.models import build_embedding_model
+
+    net = build_embedding_model("googlenet")
+    pt1 = str(tmp_path / "m.pt")
+    torch.save(net.state_dict(), pt1)
+    cm = str(tmp_path / "m.caffemodel")
+    export_main(["--to-caffemodel", pt1, cm])
+    pt2 = str(tmp_path / "m2.pt")
+    export_main(["--to-pt", cm, pt2])
+    s1 = torch.load(pt1, weights_only=True)
+    s2 = torch.load(pt2, weights_only=True)
+    # conv weights survive the round trip exactly
+    for k in s1:
+        if "conv" in k and ("weight" in k or "bias" in k):
+            torch.testing.assert_close(s1[k], s2[k])

@@ -226,3 +226,25 @@ def test_offline_recall_gpu_matches_cpu():
     gpu = recall_at_k(ft.cuda(), lt.cuda(), ks=(1, 5, 10), chunk=24)
     for k in (1, 5, 10):
         assert gpu[k] == pytest.approx(cpu[k], abs=1e-9)
+
+
+def test_loss_path_deterministic():
+    """Two identical forward+backward passes produce bit-identical loss and
+    gradients (fixed reduction orders; recall uses count-only atomics)."""
+    from npairloss_amd.config.params import NPairLossConfig
+    from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
+
+    cfg = NPairLossConfig(margin_diff=-0.05, ap_mining_region="GLOBAL",
+                          ap_mining_method="RELATIVE_HARD", identsn=-0.0,
+                          an_mining_region="LOCAL", an_mining_method="HARD")
+    f, lab = make_batch(num_classes=30, per_class=4, dim=512, seed=77)
+    outs = []
+    for _ in range(2):
+        ft = torch.from_numpy(f).float().cuda().requires_grad_(True)
+        lt = torch.from_numpy(lab).cuda()
+        out = NPairMultiClassLoss(cfg)(ft, lt)
+        out.loss.backward()
+        outs.append((out.loss.item(), out.retrieve_top1.item(), ft.grad.clone()))
+    assert outs[0][0] == outs[1][0]
+    assert outs[0][1] == outs[1][1]
+    torch.testing.assert_close(outs[0][2], outs[1][2], rtol=0, atol=0)
