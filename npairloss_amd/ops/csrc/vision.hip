@@ -226,6 +226,84 @@ __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ 
   }
 }
 
+// Wave-shuffle variant (NHWC, C % 4 == 0, 64 % (C/4) == 0, n <= 5): each
+// lane owns 4 channels; the window's out-of-quad taps come from the
+// NEIGHBOR LANES' registers via __shfl_up/__shfl_down — no LDS, no
+// barriers at all.  Pixel boundaries align with lane groups (C/4 divides
+// the 64-lane wave), so boundary lanes just zero their cross-pixel taps.
+template <typename T, bool BWD>
+__global__ void lrn_wave_kernel(const T* __restrict__ x, const T* __restrict__ dy,
+                                T* __restrict__ out, long long npix, int C,
+                                int C_q_log2, int half, float alpha_over_n,
+                                float beta, float k) {
+  const int C_q = 1 << C_q_log2;
+  const long long nquads = npix << C_q_log2;
+  for (long long q = (long long)blockIdx.x * blockDim.x + threadIdx.x; q < nquads;
+       q += (long long)gridDim.x * blockDim.x) {
+    const long long pix = q >> C_q_log2;
+    const int c0 = (int)(q & (C_q - 1)) * 4;
+    const long long base = pix * C + c0;
+    float xi[4], gi[4] = {0.f, 0.f, 0.f, 0.f};
+    ld4(x + base, xi);
+    if (BWD) ld4(dy + base, gi);
+    float xsq[4];
+#pragma unroll
+    for (int v = 0; v < 4; ++v) xsq[v] = xi[v] * xi[v];
+    // neighbor taps: left lane's top-2, right lane's bottom-2
+    const bool at_lo = (c0 == 0);
+    const bool at_hi = (c0 + 4 >= C);
+    float l2 = __shfl_up(xsq[2], 1);
+    float l3 = __shfl_up(xsq[3], 1);
+    float r0 = __shfl_down(xsq[0], 1);
+    float r1 = __shfl_down(xsq[1], 1);
+    if (at_lo) l2 = l3 = 0.f;
+    if (at_hi) r0 = r1 = 0.f;
+    const float buf[8] = {l2, l3, xsq[0], xsq[1], xsq[2], xsq[3], r0, r1};
+    float scale[4], p[4];
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      float ss = 0.f;
+#pragma unroll
+      for (int j = -2; j <= 2; ++j) {
+        const int idx = 2 + v + j;
+        // window half may be 1 (n=3): drop the +-2 taps then
+        ss += (j >= -half && j <= half) ? buf[idx] : 0.f;
+      }
+      scale[v] = k + alpha_over_n * ss;
+      p[v] = __powf(scale[v], -beta);
+    }
+    if (!BWD) {
+      float yv[4];
+#pragma unroll
+      for (int v = 0; v < 4; ++v) yv[v] = xi[v] * p[v];
+      st4(out + base, yv);
+      continue;
+    }
+    float t4[4];
+#pragma unroll
+    for (int v = 0; v < 4; ++v) t4[v] = gi[v] * (xi[v] * p[v]) / scale[v];
+    float tl2 = __shfl_up(t4[2], 1);
+    float tl3 = __shfl_up(t4[3], 1);
+    float tr0 = __shfl_down(t4[0], 1);
+    float tr1 = __shfl_down(t4[1], 1);
+    if (at_lo) tl2 = tl3 = 0.f;
+    if (at_hi) tr0 = tr1 = 0.f;
+    const float tb[8] = {tl2, tl3, t4[0], t4[1], t4[2], t4[3], tr0, tr1};
+    float dxv[4];
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      float cross = 0.f;
+#pragma unroll
+      for (int j = -2; j <= 2; ++j) {
+        const int idx = 2 + v + j;
+        cross += (j >= -half && j <= half) ? tb[idx] : 0.f;
+      }
+      dxv[v] = gi[v] * p[v] - 2.f * alpha_over_n * beta * xi[v] * cross;
+    }
+    st4(out + base, dxv);
+  }
+}
+
 // generic per-element fallback (any C): used when C > 256
 template <typename T>
 __global__ void lrn_fwd_generic(const T* __restrict__ x, T* __restrict__ y,
@@ -430,6 +508,19 @@ static int grid_for(long long total) {
     __VA_ARGS__                                                                   \
   })
 
+
+static bool wave_lrn_ok(bool nhwc, long long C, int64_t size, double k) {
+  if (!(nhwc && C % 4 == 0 && C >= 4 && C <= 256 && size <= 5 && k > 0)) return false;
+  const long long cq = C / 4;
+  return (cq & (cq - 1)) == 0 && 64 % cq == 0;
+}
+
+static int ilog2(long long v) {
+  int r = 0;
+  while ((1LL << r) < v) ++r;
+  return r;
+}
+
 torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, double k) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4);
   const bool nhwc = is_nhwc(x);
@@ -440,7 +531,17 @@ torch::Tensor lrn_fwd(torch::Tensor x, int64_t size, double alpha, double beta, 
   const long long cstride = nhwc ? 1 : S;
   auto stream = at::hip::getCurrentHIPStream();
   const float aon = (float)(alpha / size);
-  if (nhwc && C % 4 == 0 && C <= 1024 && size <= 9 && k > 0) {
+  if (wave_lrn_ok(nhwc, C, size, k)) {
+    const long long npix = B * S;
+    const long long nquads = npix * (C / 4);
+    const int blocks = (int)std::min<long long>((nquads + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 8192);
+    VISION_DISPATCH(x, "lrn_fwd", {
+      lrn_wave_kernel<T, false><<<blocks, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), nullptr,
+          reinterpret_cast<T*>(y.data_ptr()), npix, (int)C, ilog2(C / 4),
+          (int)size / 2, aon, (float)beta, (float)k);
+    });
+  } else if (nhwc && C % 4 == 0 && C <= 1024 && size <= 9 && k > 0) {
     const long long npix = B * S;
     const int ppb = NPAIR_BLOCK / (int)(C / 4);
     const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
@@ -495,7 +596,18 @@ torch::Tensor lrn_bwd(torch::Tensor x, torch::Tensor dy, int64_t size, double al
   const long long cstride = nhwc ? 1 : S;
   auto stream = at::hip::getCurrentHIPStream();
   const float aon = (float)(alpha / size);
-  if (nhwc && C % 4 == 0 && C <= 1024 && size <= 9 && k > 0) {
+  if (wave_lrn_ok(nhwc, C, size, k)) {
+    const long long npix = B * S;
+    const long long nquads = npix * (C / 4);
+    const int blocks = (int)std::min<long long>((nquads + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 8192);
+    VISION_DISPATCH(x, "lrn_bwd", {
+      lrn_wave_kernel<T, true><<<blocks, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()),
+          reinterpret_cast<const T*>(dyc.data_ptr()),
+          reinterpret_cast<T*>(dx.data_ptr()), npix, (int)C, ilog2(C / 4),
+          (int)size / 2, aon, (float)beta, (float)k);
+    });
+  } else if (nhwc && C % 4 == 0 && C <= 1024 && size <= 9 && k > 0) {
     const long long npix = B * S;
     const int ppb = NPAIR_BLOCK / (int)(C / 4);
     const int blocks = (int)std::min<long long>((npix + ppb - 1) / ppb, 8192);
