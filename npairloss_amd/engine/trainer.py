@@ -122,13 +122,14 @@ class Trainer:
         self.optimizer.step()
         self.iter += 1
         if self.divergence_check and self.iter % self.divergence_check == 0:
-            lv = float(out.loss)
+            lv = float(out.loss.detach())
             if not (lv == lv and abs(lv) != float("inf")):
                 raise FloatingPointError(
                     f"training diverged: loss={lv} at iter {self.iter}")
         return {
-            "loss": out.loss, "top1": out.retrieve_top1, "top5": out.retrieve_top5,
-            "top10": out.retrieve_top10, "asum": out.feature_asum, "lr": lr,
+            "loss": out.loss.detach(), "top1": out.retrieve_top1,
+            "top5": out.retrieve_top5, "top10": out.retrieve_top10,
+            "asum": out.feature_asum, "lr": lr,
         }
 
     # -- evaluation ---------------------------------------------------------
