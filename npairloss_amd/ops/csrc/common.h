@@ -133,6 +133,14 @@ DEVINL long long relative_index(float sn, long long size) {
   return pos;
 }
 
+// Fast x^y for x > 0: v_log_f32 + v_exp_f32 (2 hardware transcendentals).
+// hipcc lowers __powf to the full-precision OCML pow (~750 VALU
+// instructions of frexp/ldexp/special-case handling — measured as the
+// entire LRN kernel cost); this is the CUDA-__powf-equivalent fast path.
+DEVINL float fast_powf(float x, float y) {
+  return __builtin_amdgcn_exp2f(y * __builtin_amdgcn_logf(x));
+}
+
 #define HIP_CHECK_LAST()                                            \
   do {                                                              \
     hipError_t _e = hipGetLastError();                              \

@@ -110,7 +110,7 @@ __global__ void lrn_tile_kernel(const T* __restrict__ x, const T* __restrict__ d
     float ss = 0.f;
     for (int j = lo; j <= hi; ++j) ss += xsq[lds0 + j];
     const float scale = k + alpha_over_n * ss;
-    const float p = __powf(scale, -beta);     // ONE powf per element
+    const float p = fast_powf(scale, -beta);     // ONE powf per element
     if (!BWD) {
       if (act) stf(out, idx, xi * p);
       __syncthreads();
@@ -186,7 +186,7 @@ __global__ void lrn_tile4_kernel(const T* __restrict__ x, const T* __restrict__ 
 #pragma unroll
       for (int j = -half; j <= half; ++j) ss += buf[4 + v + j];  // halo zeros = clamp
       scale[v] = k + alpha_over_n * ss;
-      p[v] = __powf(scale[v], -beta);
+      p[v] = fast_powf(scale[v], -beta);
     }
     if (!BWD) {
       if (act) {
@@ -270,7 +270,7 @@ __global__ void lrn_wave_kernel(const T* __restrict__ x, const T* __restrict__ d
         ss += (j >= -half && j <= half) ? buf[idx] : 0.f;
       }
       scale[v] = k + alpha_over_n * ss;
-      p[v] = __powf(scale[v], -beta);
+      p[v] = fast_powf(scale[v], -beta);
     }
     if (!BWD) {
       float yv[4];
@@ -321,7 +321,7 @@ __global__ void lrn_fwd_generic(const T* __restrict__ x, T* __restrict__ y,
       const float v = ldf(x, base + (long long)j * cstride);
       ss += v * v;
     }
-    stf(y, i, ldf(x, i) * __powf(k + alpha_over_n * ss, -beta));
+    stf(y, i, ldf(x, i) * fast_powf(k + alpha_over_n * ss, -beta));
   }
 }
 
@@ -348,11 +348,11 @@ __global__ void lrn_bwd_generic(const T* __restrict__ x, const T* __restrict__ d
       }
       const float scale_j = k + alpha_over_n * ss;
       if (j == (int)c) scale_i = scale_j;
-      const float yj = ldf(x, base + (long long)j * cstride) * __powf(scale_j, -beta);
+      const float yj = ldf(x, base + (long long)j * cstride) * fast_powf(scale_j, -beta);
       cross += ldf(dy, base + (long long)j * cstride) * yj / scale_j;
     }
     const float xi = ldf(x, i);
-    stf(dx, i, ldf(dy, i) * __powf(scale_i, -beta) - 2.f * alpha_over_n * beta * xi * cross);
+    stf(dx, i, ldf(dy, i) * fast_powf(scale_i, -beta) - 2.f * alpha_over_n * beta * xi * cross);
   }
 }
 
