@@ -72,6 +72,7 @@ def build_trainer_from_prototxt(
     num_workers: int = 0,
     channels_last: Optional[bool] = None,
     pure_bf16: bool = False,
+    backbone: Optional[str] = None,
 ) -> Trainer:
     net = parse_net_prototxt(_clean_prototxt(net_text))
 
@@ -102,10 +103,10 @@ def build_trainer_from_prototxt(
         aug_cfg = TransformConfig.from_message(dt_layers[0].raw.get("data_transformer_l_param"))
         aug = DataTransformer(aug_cfg)
 
-    # --- backbone: the reference's conv stack is GoogLeNet v1;
+    # --- backbone: the reference's conv stack is GoogLeNet v1 (overridable);
     # L2-normalize the embedding iff the net has an L2Normalize layer
     has_l2 = bool(net.find("L2Normalize"))
-    model = build_embedding_model("googlenet", normalize=has_l2)
+    model = build_embedding_model(backbone or "googlenet", normalize=has_l2)
     if caffemodel:
         load_caffemodel_into(model, caffemodel)
 

@@ -27,6 +27,8 @@ def main(argv=None):
     p.add_argument("--pure-bf16", action="store_true",
                    help="bf16 model + fp32 master weights (replaces --amp)")
     p.add_argument("--num-workers", type=int, default=2)
+    p.add_argument("--backbone", default=None,
+                   help="override the backbone (googlenet/resnet50/vit)")
     args = p.parse_args(argv)
 
     from .config.params import SolverConfig
@@ -54,7 +56,8 @@ def main(argv=None):
     trainer = build_trainer_from_prototxt(
         net_text, solver, synthetic_classes=args.synthetic_classes,
         amp_dtype=amp_dtype, caffemodel=caffemodel,
-        num_workers=args.num_workers, pure_bf16=args.pure_bf16)
+        num_workers=args.num_workers, pure_bf16=args.pure_bf16,
+        backbone=args.backbone)
     if args.weights and args.weights.endswith(".pt"):
         trainer.restore(args.weights)
     trainer.fit(max_iter=args.max_iter)
