@@ -48,9 +48,11 @@ def parse_args():
                    help="similarity-GEMM MFMA precision on GPU")
     p.add_argument("--timers", action="store_true",
                    help="per-phase HIP-event timing report on stderr (rank 0)")
-    p.add_argument("--graph", action="store_true",
+    p.add_argument("--graph", dest="graph", action="store_true", default=None,
                    help="capture the whole train step in a hipGraph and replay "
-                        "(single GPU; removes per-kernel launch gaps)")
+                        "(default ON for single-GPU runs; removes per-kernel "
+                        "launch gaps, ~+4%%)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
     return p.parse_args()
 
 
@@ -153,9 +155,10 @@ def main():
             torch.cuda.synchronize()
 
     graph = None
-    if args.graph:
+    use_graph = args.graph if args.graph is not None else (world == 1 and use_cuda)
+    if use_graph:
         if world > 1 or not use_cuda:
-            if rank == 0:
+            if rank == 0 and args.graph:
                 print("--graph requires single-GPU CUDA; ignoring", file=sys.stderr)
         else:
             # hipGraph capture of the full step (fwd + loss + bwd + optimizer):
@@ -172,9 +175,13 @@ def main():
             torch.cuda.current_stream().wait_stream(side)
             torch.cuda.synchronize()
             timers.enabled = False  # timing events cannot record during capture
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                step(0)
+            try:
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    step(0)
+            except Exception as e:  # noqa: BLE001
+                graph = None
+                print(f"hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
 
     def run_step(i):
         if graph is not None:
