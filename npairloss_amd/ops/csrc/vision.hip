@@ -6,11 +6,14 @@
 // them with:
 //   lrn_fwd / lrn_bwd       — Caffe across-channel LRN
 //                             (scale_i = k + alpha/n * sum_{win} x_j^2,
-//                              y = x * scale^-beta).  The channel window
-//                             is shared through LDS per pixel group, so
-//                             each element costs ONE powf + ONE div (the
-//                             per-element stencil variant costs 5 of each;
-//                             torch's chain costs ~8 full-tensor passes).
+//                              y = x * scale^-beta).  Three variants by
+//                             shape: wave-shuffle (channel window taps come
+//                             from NEIGHBOR LANES' registers — no LDS, no
+//                             barriers; 4.7 TB/s), LDS pixel-group stencil,
+//                             and a generic per-element fallback.  All use
+//                             fast_powf (v_log+v_exp): hipcc's __powf is a
+//                             ~750-instruction softfloat pow and was the
+//                             entire kernel cost (profiles/kernels_r1.md).
 //   maxpool3x3_fwd / _bwd   — kernel-3 max pool (stride 1 or 2, pad 1)
 //                             storing a 1-byte argmax; backward is a
 //                             deterministic GATHER over the <=9 covering
