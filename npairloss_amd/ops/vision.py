@@ -14,14 +14,22 @@ import torch.nn.functional as F
 from . import _backend
 
 
+def _gpu_safe_dtype(x):
+    """The HIP kernels store bf16/fp32; route half through fp32."""
+    if x.is_cuda and x.dtype == torch.float16:
+        return x.float(), torch.float16
+    return x, None
+
+
 class _LRNFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, size, alpha, beta, k):
         ctx.params = (size, alpha, beta, k)
         if x.is_cuda:
-            y = _backend.ext().lrn_fwd(x, size, alpha, beta, k)
-            ctx.save_for_backward(x)
-            return y
+            xc, half = _gpu_safe_dtype(x)
+            y = _backend.ext().lrn_fwd(xc, size, alpha, beta, k)
+            ctx.save_for_backward(xc)
+            return y.half() if half else y
         ctx.save_for_backward(x)
         # CPU avg_pool3d lacks a bf16 kernel: compute in fp32, cast back
         return F.local_response_norm(x.float(), size, alpha=alpha, beta=beta, k=k).to(x.dtype)
@@ -31,7 +39,10 @@ class _LRNFn(torch.autograd.Function):
         (x,) = ctx.saved_tensors
         size, alpha, beta, k = ctx.params
         if x.is_cuda:
-            dx = _backend.ext().lrn_bwd(x, dy, size, alpha, beta, k)
+            dyc, half = _gpu_safe_dtype(dy)
+            dx = _backend.ext().lrn_bwd(x, dyc, size, alpha, beta, k)
+            if half:
+                dx = dx.half()
         else:
             with torch.enable_grad():
                 xr = x.detach().float().requires_grad_(True)
@@ -60,10 +71,11 @@ class _MaxPool3Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, stride, ceil_mode):
         if x.is_cuda:
-            y, idx = _backend.ext().maxpool3_fwd(x, stride, ceil_mode)
+            xc, half = _gpu_safe_dtype(x)
+            y, idx = _backend.ext().maxpool3_fwd(xc, stride, ceil_mode)
             ctx.save_for_backward(idx)
             ctx.meta = (stride, x.shape[2], x.shape[3], True)
-            return y
+            return y.half() if half else y
         ctx.save_for_backward(x)
         ctx.meta = (stride, x.shape[2], x.shape[3], False)
         ctx.ceil_mode = ceil_mode
@@ -74,7 +86,10 @@ class _MaxPool3Fn(torch.autograd.Function):
         (saved,) = ctx.saved_tensors
         stride, H, W, gpu = ctx.meta
         if gpu:
-            dx = _backend.ext().maxpool3_bwd(dy, saved, stride, H, W)
+            dyc, half = _gpu_safe_dtype(dy)
+            dx = _backend.ext().maxpool3_bwd(dyc, saved, stride, H, W)
+            if half:
+                dx = dx.half()
         else:
             with torch.enable_grad():
                 xr = saved.detach().requires_grad_(True)
