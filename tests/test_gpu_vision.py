@@ -70,3 +70,19 @@ def test_googlenet_gpu_matches_cpu():
     mg = m.cuda()
     y_gpu = mg(x.cuda())
     torch.testing.assert_close(y_gpu.cpu(), y_cpu, rtol=1e-3, atol=1e-4)
+
+
+def test_fp16_routes_through_fp32():
+    """half inputs upcast transparently (kernels store bf16/fp32)."""
+    x = torch.randn(2, 64, 16, 16, device="cuda", dtype=torch.float16,
+                    requires_grad=True)
+    y = CrossChannelLRN(5)(x)
+    assert y.dtype == torch.float16
+    y.sum().backward()
+    assert x.grad.dtype == torch.float16 and torch.isfinite(x.grad).all()
+    x2 = torch.randn(2, 32, 14, 14, device="cuda", dtype=torch.float16,
+                     requires_grad=True)
+    y2 = MaxPool3x3(stride=2)(x2)
+    assert y2.dtype == torch.float16
+    y2.sum().backward()
+    assert torch.isfinite(x2.grad).all()
