@@ -299,3 +299,41 @@ def test_export_cli_roundtrip(tmp_path):
     for k in s1:
         if "conv" in k and ("weight" in k or "bias" in k):
             torch.testing.assert_close(s1[k], s2[k])
+
+
+def test_trainer_test_interval_evaluation():
+    ds = SyntheticEmbeddingDataset(num_classes=16, per_class=4, dim=32, seed=0)
+    sampler = PKBatchSampler(ds.labels, identities_per_batch=8, imgs_per_identity=2, seed=0)
+    loader = torch.utils.data.DataLoader(ds, batch_sampler=sampler)
+    tds = SyntheticEmbeddingDataset(num_classes=16, per_class=4, dim=32, seed=1)
+    tsampler = PKBatchSampler(tds.labels, identities_per_batch=8, imgs_per_identity=2, seed=1)
+    tloader = torch.utils.data.DataLoader(tds, batch_sampler=tsampler)
+    from npairloss_amd.models.embedding import EmbeddingNet
+    net = EmbeddingNet(torch.nn.Linear(32, 32))
+    solver = SolverConfig(base_lr=0.01, momentum=0.9, max_iter=4, display=0,
+                          test_interval=2, test_iter=2)
+    logs = []
+    tr = Trainer(net, NPairMultiClassLoss(NPairLossConfig()), solver, loader,
+                 test_loader=tloader, device=torch.device("cpu"),
+                 log_fn=lambda m: logs.append(m))
+    tr.fit(max_iter=4)
+    assert any("TEST" in m for m in logs)
+    ev = tr.evaluate(max_batches=2)
+    assert set(ev) == {"loss", "top1", "top5", "top10"}
+
+
+def test_net_builder_with_caffemodel(tmp_path):
+    from npairloss_amd.engine.net_builder import build_trainer_from_prototxt
+    from npairloss_amd.utils.caffemodel import save_caffemodel
+
+    src = GoogLeNet()
+    with torch.no_grad():
+        src.conv1.conv.weight.fill_(0.123)
+    cm = str(tmp_path / "w.caffemodel")
+    save_caffemodel(src, cm)
+    net_text = open("examples/googlenet_npair/quickstart_net.prototxt").read()
+    solver = SolverConfig(base_lr=0.01, max_iter=1)
+    tr = build_trainer_from_prototxt(net_text, solver, device=torch.device("cpu"),
+                                     synthetic_classes=60, image_size=64,
+                                     caffemodel=cm)
+    assert (tr.model.backbone.conv1.conv.weight == 0.123).all()
