@@ -690,6 +690,8 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
   pool_strides(y, nhwc, &ysb, &ysh, &ysw, &ysc);
   const int V = (nhwc && C % 4 == 0) ? 4 : 1;
   const long long total = (long long)B * OH * OW * (C / V);
+  TORCH_CHECK(total < (1LL << 31) && (long long)B * H * W * (C / V) < (1LL << 31),
+              "maxpool3: tensor exceeds the 32-bit index fast path");
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(x, "maxpool3_fwd", {
     if (V == 4)
@@ -723,6 +725,7 @@ torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
   pool_strides(dyc, nhwc, &ysb, &ysh, &ysw, &ysc);
   const int V = (nhwc && C % 4 == 0) ? 4 : 1;
   const long long total = (long long)B * H * W * (C / V);
+  TORCH_CHECK(total < (1LL << 31), "maxpool3: tensor exceeds the 32-bit index fast path");
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(dy, "maxpool3_bwd", {
     if (V == 4)
