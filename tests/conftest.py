@@ -8,6 +8,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs an AMD GPU (run on MI355X via gpurun)")
+    config.addinivalue_line("markers", "slow: multi-process / long-running CPU tests")
 
 
 def pytest_collection_modifyitems(config, items):
