@@ -44,6 +44,10 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
 torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
                            int64_t H, int64_t W);
 
+// biasrelu.hip (EXPERIMENTAL — see the file header)
+torch::Tensor biasrelu_fwd(torch::Tensor x, torch::Tensor bias);
+std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy);
+
 // gemm_lowp.hip
 torch::Tensor sim_gemm_nt_bf16(torch::Tensor A, torch::Tensor B);
 torch::Tensor sim_gemm_nt_fp8(torch::Tensor A, torch::Tensor B);
@@ -72,6 +76,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sim_gemm_nt_bf16", &sim_gemm_nt_bf16, "bf16 MFMA similarity GEMM");
   m.def("sim_gemm_nt_fp8", &sim_gemm_nt_fp8, "fp8 e4m3 MFMA similarity GEMM");
   m.def("cast_fp8", &cast_fp8, "fp32 -> fp8 e4m3 bit pattern");
+  m.def("biasrelu_fwd", &biasrelu_fwd, "fused bias+relu forward (experimental)");
+  m.def("biasrelu_bwd", &biasrelu_bwd, "fused drelu+bias-grad backward (experimental)");
   m.def("gemm_nn", &gemm_nn, "fp32 MFMA GEMM A @ B");
   m.def("gemm_tn", &gemm_tn, "fp32 MFMA GEMM A^T @ B");
 }

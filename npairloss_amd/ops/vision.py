@@ -112,3 +112,54 @@ class MaxPool3x3(nn.Module):
 
     def extra_repr(self):
         return f"stride={self.stride}, ceil_mode={self.ceil_mode}"
+
+
+class _BiasReLUFn(torch.autograd.Function):
+    """EXPERIMENTAL (round-2 candidate, see csrc/biasrelu.hip): fused
+    y = relu(x + b[c]) with a single-pass backward producing both dx and
+    the bias gradient.  CPU path is the exact same math in torch."""
+
+    @staticmethod
+    def forward(ctx, x, bias):
+        if x.is_cuda:
+            xc, half = _gpu_safe_dtype(x)
+            y = _backend.ext().biasrelu_fwd(xc, bias.float().contiguous())
+            ctx.save_for_backward(y)
+            ctx.gpu = True
+            ctx.half = half
+            return y.half() if half else y
+        y = torch.relu(x + bias.view(1, -1, 1, 1).to(x.dtype))
+        ctx.save_for_backward(y)
+        ctx.gpu = False
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        if ctx.gpu:
+            dyc, half = _gpu_safe_dtype(dy)
+            dx, db = _backend.ext().biasrelu_bwd(y, dyc)
+            if half:
+                dx = dx.half()
+        else:
+            mask = y > 0
+            dx = dy * mask
+            db = dx.float().sum(dim=(0, 2, 3))
+        return dx, db
+
+
+class ConvBiasReLU(nn.Module):
+    """conv2d(bias=False) + fused BiasReLU.  Exposes `.weight`/`.bias` so
+    caffe_names() checkpoint mapping keeps working."""
+
+    def __init__(self, cin, cout, k, stride=1, pad=0):
+        super().__init__()
+        self.conv = nn.Conv2d(cin, cout, k, stride=stride, padding=pad, bias=False)
+        self.bias = nn.Parameter(torch.zeros(cout))
+
+    @property
+    def weight(self):
+        return self.conv.weight
+
+    def forward(self, x):
+        return _BiasReLUFn.apply(self.conv(x), self.bias)

@@ -31,6 +31,7 @@ SOURCES = [
     "gemm_f32.hip",
     "vision.hip",
     "gemm_lowp.hip",
+    "biasrelu.hip",
 ]
 
 

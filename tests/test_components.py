@@ -337,3 +337,28 @@ def test_net_builder_with_caffemodel(tmp_path):
                                      synthetic_classes=60, image_size=64,
                                      caffemodel=cm)
     assert (tr.model.backbone.conv1.conv.weight == 0.123).all()
+
+
+def test_bias_relu_cpu_parity():
+    from npairloss_amd.ops.vision import _BiasReLUFn, ConvBiasReLU
+
+    torch.manual_seed(0)
+    x = torch.randn(2, 8, 5, 7, requires_grad=True)
+    b = torch.randn(8, requires_grad=True)
+    y = _BiasReLUFn.apply(x, b)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xr = x.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    yr = torch.relu(xr + br.view(1, -1, 1, 1))
+    yr.backward(dy)
+    torch.testing.assert_close(y, yr)
+    torch.testing.assert_close(x.grad, xr.grad)
+    torch.testing.assert_close(b.grad, br.grad)
+
+    m = ConvBiasReLU(3, 4, 3, pad=1)
+    out = m(torch.randn(1, 3, 6, 6))
+    assert out.shape == (1, 4, 6, 6)
+    assert (out >= 0).all()
+    assert m.weight is m.conv.weight and m.bias.shape == (4,)
