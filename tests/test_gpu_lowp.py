@@ -77,8 +77,6 @@ def test_loss_module_sim_dtype(sim_dtype):
     from npairloss_amd.config.params import NPairLossConfig
     from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
 
-    import sys, os
-    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
     from util import make_batch
 
     cfg = NPairLossConfig(margin_diff=-0.05, ap_mining_region="GLOBAL",
