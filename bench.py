@@ -50,9 +50,14 @@ def parse_args():
                    help="per-phase HIP-event timing report on stderr (rank 0)")
     p.add_argument("--graph", dest="graph", action="store_true", default=None,
                    help="capture the whole train step in a hipGraph and replay "
-                        "(default ON for single-GPU runs; removes per-kernel "
-                        "launch gaps, ~+4%%)")
+                        "(default ON on GPU, any world size — RCCL collectives "
+                        "are captured too; falls back to eager if capture fails)")
     p.add_argument("--no-graph", dest="graph", action="store_false")
+    p.add_argument("--data-pipeline", action="store_true",
+                   help="include the input pipeline in the timed region: rotate "
+                        "8 distinct pinned-host uint8 batches through H2D copy + "
+                        "device-side DataTransformer augmentation every step "
+                        "(graph mode: copy_ into captured static tensors)")
     return p.parse_args()
 
 
@@ -121,16 +126,58 @@ def main():
                 lab[perm].to(device))
     batches = [make_batch(1000 + rank * 10 + i) for i in range(2)]
 
+    # --data-pipeline: N distinct uint8 host batches (pinned), H2D-copied into
+    # static device tensors + augmented on device INSIDE the timed region.
+    pipe = None
+    if args.data_pipeline:
+        from npairloss_amd.data.transforms import DataTransformer, TransformConfig
+
+        NPIPE = 8
+        g = torch.Generator(device="cpu").manual_seed(4321 + rank)
+        host_x, host_lab = [], []
+        for i in range(NPIPE):
+            x = torch.randint(0, 256, (B, 3, args.image, args.image),
+                              dtype=torch.uint8, generator=g)
+            x = x.to(memory_format=torch.channels_last)
+            lab = torch.arange(P).repeat_interleave(K)[:B]
+            lab = lab[torch.randperm(B, generator=g)]
+            if use_cuda:
+                x, lab = x.pin_memory(), lab.pin_memory()
+            host_x.append(x)
+            host_lab.append(lab)
+        static_u8 = host_x[0].to(device, non_blocking=True)
+        static_lab = host_lab[0].to(device, non_blocking=True)
+        aug = DataTransformer(TransformConfig(
+            rotate_angle_scope=0.18, translation_w_scope=8.0,
+            translation_h_scope=8.0, scale_w_scope=1.1, scale_h_scope=1.1,
+            h_flip=True))
+        pipe = dict(host_x=host_x, host_lab=host_lab, static_u8=static_u8,
+                    static_lab=static_lab, aug=aug, n=NPIPE)
+
     last_out = {}
     from npairloss_amd.utils.profiling import PhaseTimers
     timers = PhaseTimers(enabled=args.timers, use_cuda=use_cuda)
 
+    def feed(i):
+        """H2D copy of batch i into the static tensors (outside any graph)."""
+        if pipe is not None:
+            j = i % pipe["n"]
+            pipe["static_u8"].copy_(pipe["host_x"][j], non_blocking=True)
+            pipe["static_lab"].copy_(pipe["host_lab"][j], non_blocking=True)
+
     def step(i):
         nonlocal last_out
-        x, lab = batches[i % 2]
+        if pipe is not None:
+            with timers.phase("data"):
+                # decode + scale + device-side random affine augmentation
+                x = pipe["static_u8"].float().sub_(127.5).mul_(1.0 / 64.0)
+                x = pipe["aug"](x).to(memory_format=torch.channels_last)
+                lab = pipe["static_lab"]
+        else:
+            x, lab = batches[i % 2]
         if args.pure_bf16:
             x = x.to(torch.bfloat16)
-        opt.zero_grad(set_to_none=True)
+        reducer.zero_grad()
         with timers.phase("forward"):
             if amp:
                 with torch.autocast("cuda", dtype=torch.bfloat16):
@@ -155,35 +202,45 @@ def main():
             torch.cuda.synchronize()
 
     graph = None
-    use_graph = args.graph if args.graph is not None else (world == 1 and use_cuda)
+    use_graph = args.graph if args.graph is not None else use_cuda
+    if use_graph and not use_cuda:
+        if rank == 0 and args.graph:
+            print("--graph requires CUDA; ignoring", file=sys.stderr)
+        use_graph = False
     if use_graph:
-        if world > 1 or not use_cuda:
-            if rank == 0 and args.graph:
-                print("--graph requires single-GPU CUDA; ignoring", file=sys.stderr)
-        else:
-            # hipGraph capture of the full step (fwd + loss + bwd + optimizer):
-            # launch-bound inner loop becomes one graph replay.  Inputs stay
-            # at fixed addresses (batches[0]); MIOpen find must have run
-            # during warmup (cudnn.benchmark caches per shape).
-            for i in range(max(args.warmup, 3)):
+        # hipGraph capture of the full step (fwd + loss + bwd + comm +
+        # optimizer): the launch-bound inner loop becomes one graph replay.
+        # Inputs stay at fixed addresses (batches[0] / the static pipeline
+        # tensors); gradients live in the reducer's persistent flat buckets,
+        # so the RCCL all-reduces capture with fixed buffers at world>1.
+        # MIOpen find must have run during warmup (cudnn.benchmark per shape).
+        for i in range(max(args.warmup, 3)):
+            feed(i)
+            step(0)
+        torch.cuda.synchronize()
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()  # align ranks so every rank captures the same comm
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            step(0)  # warm the allocator on the capture stream
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        timers.enabled = False  # timing events cannot record during capture
+        try:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
                 step(0)
-            torch.cuda.synchronize()
-            side = torch.cuda.Stream()
-            side.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(side):
-                step(0)  # warm the allocator on the capture stream
-            torch.cuda.current_stream().wait_stream(side)
-            torch.cuda.synchronize()
-            timers.enabled = False  # timing events cannot record during capture
-            try:
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
-                    step(0)
-            except Exception as e:  # noqa: BLE001
-                graph = None
-                print(f"hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
+        except Exception as e:  # noqa: BLE001
+            graph = None
+            print(f"hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
 
     def run_step(i):
+        feed(i)
         if graph is not None:
             graph.replay()
         else:
@@ -226,7 +283,7 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16" if (amp or args.pure_bf16) else "fp32",
-            "data": "synthetic",
+            "data": "synthetic+pipeline" if pipe is not None else "synthetic",
             "config": {
                 "model": args.model,
                 "global_batch": world * B,
@@ -237,6 +294,7 @@ def main():
                 "sim_dtype": args.sim_dtype,
                 "parallelism": f"dp{world}",
                 "hip_graph": graph is not None,
+                "data_pipeline": pipe is not None,
                 "recall_top1_last_step": float(last_out["top1"]) if last_out else None,
             },
         }

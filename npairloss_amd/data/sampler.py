@@ -32,6 +32,7 @@ class PKBatchSampler:
         self.K = imgs_per_identity
         self.shuffle = shuffle
         self.rand_identity = rand_identity
+        self.base_seed = seed
         self.rng = random.Random(seed)
         self.by_class: Dict[int, List[int]] = defaultdict(list)
         for idx, lab in enumerate(self.labels):
@@ -47,7 +48,10 @@ class PKBatchSampler:
         )
 
     def set_epoch(self, epoch: int) -> None:
-        self.rng = random.Random(hash((epoch, id(self) & 0xFFFF)) & 0x7FFFFFFF)
+        """Reseed deterministically from (base_seed, epoch) so runs are
+        reproducible; base_seed should already be rank-folded (the builder
+        passes seed = random_seed + rank so ranks draw disjoint batches)."""
+        self.rng = random.Random(self.base_seed * 1_000_003 + epoch)
 
     def __len__(self) -> int:
         return self.batches_per_epoch

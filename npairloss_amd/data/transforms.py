@@ -53,7 +53,12 @@ class DataTransformer:
         self.generator = generator
 
     def _rand(self, n, device, lo, hi):
-        u = torch.rand(n, device="cpu", generator=self.generator).to(device)
+        if self.generator is not None:
+            u = torch.rand(n, device="cpu", generator=self.generator).to(device)
+        else:
+            # device-side RNG: no H2D sync, and hipGraph-capture-safe (the
+            # philox offset advances per replay so augmentations stay random)
+            u = torch.rand(n, device=device)
         return lo + (hi - lo) * u
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:

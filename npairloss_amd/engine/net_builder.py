@@ -118,7 +118,17 @@ def build_trainer_from_prototxt(
         lcfg = NPairLossConfig()
     loss_mod = NPairMultiClassLoss(lcfg)
 
-    # --- dataset: real list-file folder when it exists, else synthetic
+    # --- dataset: real list-file folder when it exists, else synthetic.
+    # Seeds are rank-folded (solver.random_seed + rank): with the loss
+    # all-gathering embeddings across ranks, identical per-rank batches would
+    # inject world_size duplicates of every sample into the global batch —
+    # perfect sim~1.0 "positives" that pair_is_self does not exclude,
+    # corrupting mining and Recall@k.  Distinct seeds keep ranks drawing
+    # independent batches like the reference's per-rank MPI data layers.
+    from ..parallel import collectives as comm
+
+    base_seed = solver.random_seed if solver.random_seed is not None else 0
+    rank_seed = base_seed + comm.rank()
     use_real = spec.root_folder and os.path.isdir(spec.root_folder) and os.path.isfile(spec.source)
     if use_real:
         from ..data.folder import FolderListDataset
@@ -126,21 +136,25 @@ def build_trainer_from_prototxt(
         ds = FolderListDataset(spec.root_folder, spec.source,
                                new_height=spec.new_height, new_width=spec.new_width)
     else:
+        # rank-folded dataset seed: the synthetic pool is small, so index
+        # collisions across ranks would be frequent — rank-unique images
+        # (with the shared label space) avoid exact-duplicate embeddings
         n_cls = max(synthetic_classes, spec.identities_per_batch)
         ds = SyntheticImageDataset(num_classes=n_cls,
                                    per_class=max(synthetic_per_class, spec.imgs_per_identity),
-                                   image_size=img, seed=0)
+                                   image_size=img, seed=base_seed * 131 + comm.rank())
     sampler = PKBatchSampler(ds.labels, spec.identities_per_batch, spec.imgs_per_identity,
-                             shuffle=spec.shuffle, rand_identity=spec.rand_identity, seed=0)
+                             shuffle=spec.shuffle, rand_identity=spec.rand_identity,
+                             seed=rank_seed)
     loader = torch.utils.data.DataLoader(ds, batch_sampler=sampler, num_workers=num_workers)
 
     test_loader = None
     if test_spec is not None:
         tds = SyntheticImageDataset(num_classes=max(synthetic_classes, test_spec.identities_per_batch),
                                     per_class=max(synthetic_per_class, test_spec.imgs_per_identity),
-                                    image_size=img, seed=1)
+                                    image_size=img, seed=(base_seed + 1) * 131 + comm.rank())
         tsampler = PKBatchSampler(tds.labels, test_spec.identities_per_batch,
-                                  test_spec.imgs_per_identity, seed=1)
+                                  test_spec.imgs_per_identity, seed=rank_seed + 7919)
         test_loader = torch.utils.data.DataLoader(tds, batch_sampler=tsampler, num_workers=num_workers)
 
     if channels_last is None:
@@ -148,5 +162,9 @@ def build_trainer_from_prototxt(
     trainer = Trainer(model, loss_mod, solver, loader, test_loader=test_loader,
                       device=device, amp_dtype=amp_dtype,
                       channels_last=channels_last, pure_bf16=pure_bf16)
-    trainer.augment = aug  # applied by train CLI before the model (TRAIN phase)
+    trainer.augment = aug  # applied in train_step before the model (TRAIN phase)
+    # Caffe transform_param (mean subtraction / crop) applies to real 0-255
+    # image sources; synthetic data is already zero-mean at crop size.
+    has_tp = bool(data_layers) and data_layers[0].raw.has("transform_param")
+    trainer.preprocess = tcfg if (use_real and has_tp) else None
     return trainer

@@ -69,6 +69,11 @@ class Trainer:
         self.avg = AverageWindow(solver.average_loss)
         self.history = []
         self.augment = None  # optional TRAIN-phase batch augmentation (DataTransformer)
+        # optional Caffe transform_param (TransformConfig): mean subtraction +
+        # center crop applied to every batch BEFORE augmentation, matching the
+        # reference data layer (usage/def.prototxt:10-16).  Set by the net
+        # builder when training from a real 0-255 image source.
+        self.preprocess = None
         # failure detection: raise on NaN/Inf loss every N iters (0 = off).
         # (the reference had none — an NaN would silently poison the run)
         self.divergence_check = 25
@@ -100,6 +105,9 @@ class Trainer:
     def train_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
         self.model.train()
         images = images.to(self.device, non_blocking=True)
+        if self.preprocess is not None and images.dim() == 4:
+            from ..data.transforms import preprocess as _pp
+            images = _pp(images, self.preprocess)
         if self.augment is not None and images.dim() == 4:
             images = self.augment(images)
         if self.channels_last and images.dim() == 4:
@@ -109,7 +117,7 @@ class Trainer:
             images = images.to(torch.bfloat16)
         lr = self.solver.lr_at(self.iter)
         self.optimizer.set_lr(lr)
-        self.optimizer.zero_grad(set_to_none=True)
+        self.reducer.zero_grad()  # zeroes the persistent flat grad buckets
         if self.amp_dtype is not None and self.device.type == "cuda":
             with torch.autocast("cuda", dtype=self.amp_dtype):
                 feats = self.model(images)
@@ -143,6 +151,9 @@ class Trainer:
         acc = collections.defaultdict(float)
         for images, labels in self.test_loader:
             images = images.to(self.device, non_blocking=True)
+            if self.preprocess is not None and images.dim() == 4:
+                from ..data.transforms import preprocess as _pp
+                images = _pp(images, self.preprocess)
             labels = labels.to(self.device, non_blocking=True)
             feats = self.model(images)
             out = self.loss(feats, labels)
@@ -160,11 +171,16 @@ class Trainer:
         max_iter = max_iter or self.solver.max_iter
         t0 = time.time()
         it_timer = time.time()
+        epoch = 0
         data_iter = iter(self.train_loader)
         while self.iter < max_iter:
             try:
                 images, labels = next(data_iter)
             except StopIteration:
+                epoch += 1
+                bs = getattr(self.train_loader, "batch_sampler", None)
+                if bs is not None and hasattr(bs, "set_epoch"):
+                    bs.set_epoch(epoch)  # deterministic (base_seed, epoch) reseed
                 data_iter = iter(self.train_loader)
                 images, labels = next(data_iter)
             stats = self.train_step(images, labels)

@@ -17,7 +17,7 @@ from typing import Dict, Tuple
 import torch
 from torch import nn
 
-from ..ops.vision import CrossChannelLRN, MaxPool3x3
+from ..ops.vision import ConvBiasReLU, CrossChannelLRN, MaxPool3x3
 
 
 class ConvReLU(nn.Module):
@@ -30,18 +30,29 @@ class ConvReLU(nn.Module):
         return self.relu(self.conv(x))
 
 
+def _conv_block(cin, cout, k, stride=1, pad=0, fused=True):
+    """conv+bias+relu unit.  fused=True (default) keeps bias OUT of the conv
+    and runs the fused BiasReLU kernel: backward collapses torch's
+    threshold_backward + generic bias-grad reduce into one pass
+    (csrc/biasrelu.hip); fused=False is the plain conv(bias)+ReLU pair."""
+    if fused:
+        return ConvBiasReLU(cin, cout, k, stride=stride, pad=pad)
+    return ConvReLU(cin, cout, k, stride=stride, pad=pad)
+
+
 class Inception(nn.Module):
     """The 4-branch inception block: 1x1 | 1x1->3x3 | 1x1->5x5 | pool->1x1."""
 
-    def __init__(self, cin: int, c1: int, c3r: int, c3: int, c5r: int, c5: int, cp: int):
+    def __init__(self, cin: int, c1: int, c3r: int, c3: int, c5r: int, c5: int, cp: int,
+                 fused: bool = True):
         super().__init__()
-        self.b1 = ConvReLU(cin, c1, 1)
-        self.b3_reduce = ConvReLU(cin, c3r, 1)
-        self.b3 = ConvReLU(c3r, c3, 3, pad=1)
-        self.b5_reduce = ConvReLU(cin, c5r, 1)
-        self.b5 = ConvReLU(c5r, c5, 5, pad=2)
+        self.b1 = _conv_block(cin, c1, 1, fused=fused)
+        self.b3_reduce = _conv_block(cin, c3r, 1, fused=fused)
+        self.b3 = _conv_block(c3r, c3, 3, pad=1, fused=fused)
+        self.b5_reduce = _conv_block(cin, c5r, 1, fused=fused)
+        self.b5 = _conv_block(c5r, c5, 5, pad=2, fused=fused)
         self.pool = MaxPool3x3(stride=1)  # fused gfx950 kernel on GPU
-        self.pool_proj = ConvReLU(cin, cp, 1)
+        self.pool_proj = _conv_block(cin, cp, 1, fused=fused)
 
     def forward(self, x):
         return torch.cat(
@@ -71,17 +82,19 @@ class GoogLeNet(nn.Module):
 
     embed_dim = 1024
 
-    def __init__(self, dropout: float = 0.4):
+    def __init__(self, dropout: float = 0.4, fused_bias_relu: bool = True):
         super().__init__()
-        self.conv1 = ConvReLU(3, 64, 7, stride=2, pad=3)
+        fused = fused_bias_relu
+        self.conv1 = _conv_block(3, 64, 7, stride=2, pad=3, fused=fused)
         self.pool1 = MaxPool3x3(stride=2)
         self.norm1 = CrossChannelLRN(5, alpha=1e-4, beta=0.75)
-        self.conv2_reduce = ConvReLU(64, 64, 1)
-        self.conv2 = ConvReLU(64, 192, 3, pad=1)
+        self.conv2_reduce = _conv_block(64, 64, 1, fused=fused)
+        self.conv2 = _conv_block(64, 192, 3, pad=1, fused=fused)
         self.norm2 = CrossChannelLRN(5, alpha=1e-4, beta=0.75)
         self.pool2 = MaxPool3x3(stride=2)
         self.inception = nn.ModuleDict({
-            name: Inception(_INCEPTION_IN[name], *cfg) for name, cfg in _INCEPTION_CFG.items()
+            name: Inception(_INCEPTION_IN[name], *cfg, fused=fused)
+            for name, cfg in _INCEPTION_CFG.items()
         })
         self.pool3 = MaxPool3x3(stride=2)
         self.pool4 = MaxPool3x3(stride=2)
@@ -95,6 +108,8 @@ class GoogLeNet(nn.Module):
                 nn.init.xavier_uniform_(m.weight)
                 if m.bias is not None:
                     nn.init.constant_(m.bias, 0.2)  # def.prototxt:109-112 filler
+            elif isinstance(m, ConvBiasReLU):
+                nn.init.constant_(m.bias, 0.2)  # same filler, bias lives outside conv
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = self.norm1(self.pool1(self.conv1(x)))
@@ -113,12 +128,17 @@ class GoogLeNet(nn.Module):
 
     # -- Caffe name mapping (for .caffemodel loading) ------------------------
 
-    def caffe_names(self) -> Dict[str, nn.Conv2d]:
-        """Caffe layer name -> conv module holding its (weight, bias)."""
-        m: Dict[str, nn.Conv2d] = {
-            "conv1/7x7_s2": self.conv1.conv,
-            "conv2/3x3_reduce": self.conv2_reduce.conv,
-            "conv2/3x3": self.conv2.conv,
+    def caffe_names(self) -> Dict[str, nn.Module]:
+        """Caffe layer name -> module holding its (weight, bias).  For
+        ConvBiasReLU the module itself carries .weight (property into the
+        bias-free conv) + .bias, so checkpoint IO is layout-agnostic."""
+        def unit(mod):
+            return mod.conv if isinstance(mod, ConvReLU) else mod
+
+        m: Dict[str, nn.Module] = {
+            "conv1/7x7_s2": unit(self.conv1),
+            "conv2/3x3_reduce": unit(self.conv2_reduce),
+            "conv2/3x3": unit(self.conv2),
         }
         branch_names = {
             "b1": "1x1",
@@ -130,5 +150,5 @@ class GoogLeNet(nn.Module):
         }
         for blk, mod in self.inception.items():
             for attr, suffix in branch_names.items():
-                m[f"inception_{blk}/{suffix}"] = getattr(mod, attr).conv
+                m[f"inception_{blk}/{suffix}"] = unit(getattr(mod, attr))
         return m

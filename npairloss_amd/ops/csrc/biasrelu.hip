@@ -6,17 +6,24 @@
 // activation tensor — torch's threshold_backward (~0.9 ms/step) plus the
 // generic NHWC bias-grad reduce (~1.4 ms/step).  Moving bias out of the
 // conv and fusing relu+bias forward / drelu+bias-grad backward folds the
-// backward to ONE pass with a deterministic two-stage column reduce.
+// backward to ONE pass.
 //
-// Status: EXPERIMENTAL this round — compile-verified + CPU-parity-tested;
-// GPU numerics tests are gated behind NPAIR_EXPERIMENTAL=1
-// (tests/test_gpu_experimental.py) pending GPU validation next round.
-// Nothing routes through these kernels unless GoogLeNet is built with
-// fused_bias_relu=True.
+// NHWC fast path (C % VEC == 0, VEC = 16B/sizeof(T)): 16-byte vector
+// loads/stores, and a FIXED-CHANNEL schedule — the host picks the grid so
+// the grid-stride (in VEC-element groups) is a multiple of C/VEC, which
+// pins every thread to the same VEC-channel window for its whole loop.
+// Bias loads hoist into registers (forward) and the bias gradient
+// accumulates in registers with an ordered LDS block reduce + ordered
+// cross-block finalize (backward) — fully DETERMINISTIC, no atomics.
+// The NCHW / odd-C fallback uses LDS atomicAdd partials (deterministic
+// only up to float addition order; nothing in the framework routes
+// 4D activations through it — channels_last is the default layout).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <hip/hip_bf16.h>
+
+#include <numeric>
 
 #include "common.h"
 
@@ -31,8 +38,101 @@ template <> DEVINL void stf1<__hip_bfloat16>(__hip_bfloat16* p, long long i, flo
   p[i] = __float2bfloat16(v);
 }
 
+template <typename T> DEVINL float cvt_to_f(T v);
+template <> DEVINL float cvt_to_f<float>(float v) { return v; }
+template <> DEVINL float cvt_to_f<__hip_bfloat16>(__hip_bfloat16 v) { return __bfloat162float(v); }
+template <typename T> DEVINL T cvt_from_f(float v);
+template <> DEVINL float cvt_from_f<float>(float v) { return v; }
+template <> DEVINL __hip_bfloat16 cvt_from_f<__hip_bfloat16>(float v) { return __float2bfloat16(v); }
+
+// 16-byte vector of VEC = 16/sizeof(T) elements
+template <typename T>
+struct alignas(16) Pack16 {
+  static constexpr int N = 16 / sizeof(T);
+  T v[N];
+};
+
 // ---------------------------------------------------------------------------
-// forward: y = relu(x + b[c]); NHWC (cstride 1) or NCHW (cstride S)
+// NHWC fast path: vectorized + fixed-channel schedule
+// ---------------------------------------------------------------------------
+
+// forward: y = relu(x + b[c]); host guarantees (gridDim*blockDim) % (C/VEC)==0
+template <typename T>
+__global__ void biasrelu_fwd_vec_kernel(const T* __restrict__ x,
+                                        const float* __restrict__ bias,
+                                        T* __restrict__ y, long long groups,
+                                        int C) {
+  constexpr int V = Pack16<T>::N;
+  const long long t0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int c0 = (int)(((long long)V * t0) % C);  // fixed for all iterations
+  float b[V];
+#pragma unroll
+  for (int k = 0; k < V; ++k) b[k] = bias[c0 + k];
+  const Pack16<T>* xp = reinterpret_cast<const Pack16<T>*>(x);
+  Pack16<T>* yp = reinterpret_cast<Pack16<T>*>(y);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long g = t0; g < groups; g += stride) {
+    Pack16<T> xv = xp[g];
+    Pack16<T> yv;
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      const float v = cvt_to_f(xv.v[k]) + b[k];
+      yv.v[k] = cvt_from_f<T>(v > 0.f ? v : 0.f);
+    }
+    yp[g] = yv;
+  }
+}
+
+// backward: dx = dy * (y > 0); db[c] = ordered sum of dx over sites.
+// Each thread owns channels [c0, c0+V); register accumulate, then an
+// ordered per-block LDS reduce, then the ordered cross-block finalize.
+template <typename T>
+__global__ void biasrelu_bwd_vec_kernel(const T* __restrict__ y,
+                                        const T* __restrict__ dy,
+                                        T* __restrict__ dx,
+                                        float* __restrict__ partials,
+                                        long long groups, int C) {
+  constexpr int V = Pack16<T>::N;
+  extern __shared__ float lds[];  // blockDim * V floats
+  const long long t0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int c0 = (int)(((long long)V * t0) % C);
+  float acc[V];
+#pragma unroll
+  for (int k = 0; k < V; ++k) acc[k] = 0.f;
+  const Pack16<T>* yp = reinterpret_cast<const Pack16<T>*>(y);
+  const Pack16<T>* dyp = reinterpret_cast<const Pack16<T>*>(dy);
+  Pack16<T>* dxp = reinterpret_cast<Pack16<T>*>(dx);
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long g = t0; g < groups; g += stride) {
+    Pack16<T> yv = yp[g];
+    Pack16<T> dyv = dyp[g];
+    Pack16<T> dxv;
+#pragma unroll
+    for (int k = 0; k < V; ++k) {
+      const float gk = (cvt_to_f(yv.v[k]) > 0.f) ? cvt_to_f(dyv.v[k]) : 0.f;
+      dxv.v[k] = cvt_from_f<T>(gk);
+      acc[k] += gk;
+    }
+    dxp[g] = dxv;
+  }
+#pragma unroll
+  for (int k = 0; k < V; ++k) lds[threadIdx.x * V + k] = acc[k];
+  __syncthreads();
+  // ordered per-channel sum over the block's threads (ascending thread id)
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float s = 0.f;
+    const long long base = (long long)blockIdx.x * blockDim.x;
+    for (int t = 0; t < blockDim.x; ++t) {
+      const int tc0 = (int)(((long long)V * (base + t)) % C);
+      const int off = c - tc0;
+      if (off >= 0 && off < V) s += lds[t * V + off];
+    }
+    partials[(long long)blockIdx.x * C + c] = s;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scalar fallback (NCHW or C not a multiple of VEC)
 // ---------------------------------------------------------------------------
 
 template <typename T>
@@ -47,12 +147,6 @@ __global__ void biasrelu_fwd_kernel(const T* __restrict__ x,
     stf1(y, i, v > 0.f ? v : 0.f);
   }
 }
-
-// ---------------------------------------------------------------------------
-// backward: dx = dy * (y > 0); db[c] = sum over sites of dx
-// Stage 1: per-block LDS bins -> partials[block][C]; stage 2 sums blocks.
-// Deterministic (fixed block count, ordered stage-2 sum).
-// ---------------------------------------------------------------------------
 
 template <typename T>
 __global__ void biasrelu_bwd_kernel(const T* __restrict__ y,
@@ -75,6 +169,7 @@ __global__ void biasrelu_bwd_kernel(const T* __restrict__ y,
   for (int c = threadIdx.x; c < C; c += blockDim.x) out[c] = bins[c];
 }
 
+// ordered over blocks (ascending) — deterministic stage 2 for both paths
 __global__ void biasrelu_db_finalize_kernel(const float* __restrict__ partials,
                                             float* __restrict__ db, int C,
                                             int nblocks) {
@@ -94,6 +189,16 @@ static bool br_nhwc(const torch::Tensor& t) {
   return t.is_contiguous(at::MemoryFormat::ChannelsLast);
 }
 
+// grid size such that (grid * NPAIR_BLOCK) % q == 0 (q = C/VEC): pins every
+// thread's channel window across grid-stride iterations
+static int br_fixed_grid(long long groups, int q) {
+  long long desired = (groups + NPAIR_BLOCK * 4 - 1) / (NPAIR_BLOCK * 4);
+  if (desired < 256) desired = 256;
+  if (desired > 4096) desired = 4096;
+  const int m = q / std::gcd((long long)q, (long long)NPAIR_BLOCK);
+  return (int)((desired + m - 1) / m * m);
+}
+
 torch::Tensor biasrelu_fwd(torch::Tensor x, torch::Tensor bias) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4);
   TORCH_CHECK(bias.is_cuda() && bias.dtype() == torch::kFloat32 && bias.is_contiguous());
@@ -103,16 +208,25 @@ torch::Tensor biasrelu_fwd(torch::Tensor x, torch::Tensor bias) {
   const long long C = x.size(1), S = x.size(2) * x.size(3);
   const long long total = x.size(0) * C * S;
   const long long cstride = nhwc ? 1 : S;
-  const int blocks = (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 4096);
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
       x.scalar_type(), "biasrelu_fwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
     TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
                 "biasrelu: bf16/fp32 only");
-    biasrelu_fwd_kernel<T><<<blocks, NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const T*>(xc.data_ptr()), bias.data_ptr<float>(),
-        reinterpret_cast<T*>(y.data_ptr()), total, (int)C, cstride);
+    constexpr int V = Pack16<T>::N;
+    if (nhwc && C % V == 0) {
+      const long long groups = total / V;
+      const int grid = br_fixed_grid(groups, (int)(C / V));
+      biasrelu_fwd_vec_kernel<T><<<grid, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), bias.data_ptr<float>(),
+          reinterpret_cast<T*>(y.data_ptr()), groups, (int)C);
+    } else {
+      const int blocks = (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 4096);
+      biasrelu_fwd_kernel<T><<<blocks, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), bias.data_ptr<float>(),
+          reinterpret_cast<T*>(y.data_ptr()), total, (int)C, cstride);
+    }
   });
   HIP_CHECK_LAST();
   return y;
@@ -128,8 +242,6 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
   const long long C = y.size(1), S = y.size(2) * y.size(3);
   const long long total = y.size(0) * C * S;
   const long long cstride = nhwc ? 1 : S;
-  const int blocks = (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 1024);
-  auto partials = torch::empty({blocks, C}, y.options().dtype(torch::kFloat32));
   auto db = torch::empty({C}, y.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
@@ -137,14 +249,31 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
     TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
                 "biasrelu: bf16/fp32 only");
-    biasrelu_bwd_kernel<T><<<blocks, NPAIR_BLOCK, (size_t)C * sizeof(float), stream>>>(
-        reinterpret_cast<const T*>(yc.data_ptr()),
-        reinterpret_cast<const T*>(dyc.data_ptr()),
-        reinterpret_cast<T*>(dx.data_ptr()), partials.data_ptr<float>(),
-        total, (int)C, cstride);
+    constexpr int V = Pack16<T>::N;
+    if (nhwc && C % V == 0) {
+      const long long groups = total / V;
+      const int grid = br_fixed_grid(groups, (int)(C / V));
+      auto partials = torch::empty({grid, C}, y.options().dtype(torch::kFloat32));
+      biasrelu_bwd_vec_kernel<T><<<grid, NPAIR_BLOCK,
+                                   (size_t)NPAIR_BLOCK * V * sizeof(float), stream>>>(
+          reinterpret_cast<const T*>(yc.data_ptr()),
+          reinterpret_cast<const T*>(dyc.data_ptr()),
+          reinterpret_cast<T*>(dx.data_ptr()), partials.data_ptr<float>(),
+          groups, (int)C);
+      biasrelu_db_finalize_kernel<<<(int)std::min<long long>((C + 255) / 256, 64), 256, 0, stream>>>(
+          partials.data_ptr<float>(), db.data_ptr<float>(), (int)C, grid);
+    } else {
+      const int blocks = (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 1024);
+      auto partials = torch::empty({blocks, C}, y.options().dtype(torch::kFloat32));
+      biasrelu_bwd_kernel<T><<<blocks, NPAIR_BLOCK, (size_t)C * sizeof(float), stream>>>(
+          reinterpret_cast<const T*>(yc.data_ptr()),
+          reinterpret_cast<const T*>(dyc.data_ptr()),
+          reinterpret_cast<T*>(dx.data_ptr()), partials.data_ptr<float>(),
+          total, (int)C, cstride);
+      biasrelu_db_finalize_kernel<<<(int)std::min<long long>((C + 255) / 256, 64), 256, 0, stream>>>(
+          partials.data_ptr<float>(), db.data_ptr<float>(), (int)C, blocks);
+    }
   });
-  biasrelu_db_finalize_kernel<<<(int)std::min<long long>((C + 255) / 256, 64), 256, 0, stream>>>(
-      partials.data_ptr<float>(), db.data_ptr<float>(), (int)C, blocks);
   HIP_CHECK_LAST();
   return {dx, db};
 }

@@ -115,12 +115,13 @@ class MaxPool3x3(nn.Module):
 
 
 class _BiasReLUFn(torch.autograd.Function):
-    """EXPERIMENTAL (round-2 candidate, see csrc/biasrelu.hip): fused
-    y = relu(x + b[c]) with a single-pass backward producing both dx and
-    the bias gradient.  CPU path is the exact same math in torch."""
+    """Fused y = relu(x + b[c]) with a single-pass backward producing both
+    dx and the bias gradient (csrc/biasrelu.hip — vectorized deterministic
+    NHWC kernels).  CPU path is the exact same math in torch."""
 
     @staticmethod
     def forward(ctx, x, bias):
+        ctx.bias_dtype = bias.dtype
         if x.is_cuda:
             xc, half = _gpu_safe_dtype(x)
             y = _backend.ext().biasrelu_fwd(xc, bias.float().contiguous())
@@ -145,7 +146,7 @@ class _BiasReLUFn(torch.autograd.Function):
             mask = y > 0
             dx = dy * mask
             db = dx.float().sum(dim=(0, 2, 3))
-        return dx, db
+        return dx, db.to(ctx.bias_dtype)
 
 
 class ConvBiasReLU(nn.Module):
