@@ -68,7 +68,11 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     use_cuda = torch.cuda.is_available()
-    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    # modulo lets a multi-rank smoke run on fewer GPUs than ranks (e.g. the
+    # 2-rank RCCL path check on a 1-GPU box); production launches have
+    # local_rank < device_count so this is the identity there
+    dev_idx = local_rank % max(1, torch.cuda.device_count()) if use_cuda else 0
+    device = torch.device(f"cuda:{dev_idx}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
         torch.backends.cudnn.benchmark = True  # MIOpen find-best for fixed shapes
