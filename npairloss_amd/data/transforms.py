@@ -72,8 +72,9 @@ class DataTransformer:
         sw = self._sample_scale(B, dev, cfg.scale_w_scope)
         sh = self._sample_scale(B, dev, cfg.scale_h_scope)
         if cfg.h_flip:
-            flip = torch.where(self._rand(B, dev, 0.0, 1.0) < 0.5,
-                               torch.tensor(-1.0, device=dev), torch.tensor(1.0, device=dev))
+            # arithmetic form (no host-constant tensors: H2D copies of
+            # pageable memory are forbidden during hipGraph capture)
+            flip = 1.0 - 2.0 * (self._rand(B, dev, 0.0, 1.0) < 0.5).float()
         else:
             flip = torch.ones(B, device=dev)
         cos, sin = torch.cos(ang), torch.sin(ang)
