@@ -118,16 +118,25 @@ __global__ void biasrelu_bwd_vec_kernel(const T* __restrict__ y,
 #pragma unroll
   for (int k = 0; k < V; ++k) lds[threadIdx.x * V + k] = acc[k];
   __syncthreads();
-  // ordered per-channel sum over the block's threads (ascending thread id)
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float s = 0.f;
-    const long long base = (long long)blockIdx.x * blockDim.x;
-    for (int t = 0; t < blockDim.x; ++t) {
-      const int tc0 = (int)(((long long)V * (base + t)) % C);
-      const int off = c - tc0;
-      if (off >= 0 && off < V) s += lds[t * V + off];
+  // Ordered per-channel-window sum over the block's threads.  Thread t's
+  // window index is (base + t) mod q with q = C/V, so the threads sharing
+  // window j are t0, t0+q, t0+2q, ... (ascending — deterministic order);
+  // host guarantees q <= blockDim.  No modulo in the loop body.
+  const int q = C / V;
+  const int base_mod = (int)(((long long)blockIdx.x * blockDim.x) % q);
+  for (int j = threadIdx.x; j < q; j += blockDim.x) {
+    int t0 = j - base_mod;
+    if (t0 < 0) t0 += q;
+    float s[V];
+#pragma unroll
+    for (int k = 0; k < V; ++k) s[k] = 0.f;
+    for (int t = t0; t < (int)blockDim.x; t += q) {
+#pragma unroll
+      for (int k = 0; k < V; ++k) s[k] += lds[t * V + k];
     }
-    partials[(long long)blockIdx.x * C + c] = s;
+    float* out = partials + (long long)blockIdx.x * C + (long long)j * V;
+#pragma unroll
+    for (int k = 0; k < V; ++k) out[k] = s[k];
   }
 }
 
@@ -169,7 +178,22 @@ __global__ void biasrelu_bwd_kernel(const T* __restrict__ y,
   for (int c = threadIdx.x; c < C; c += blockDim.x) out[c] = bins[c];
 }
 
-// ordered over blocks (ascending) — deterministic stage 2 for both paths
+// ordered over blocks (ascending) — deterministic stage 2 for both paths.
+// Split into fixed chunks along the block axis (blockIdx.y) so the serial
+// depth stays short; each chunk sums ascending, then the chunk results sum
+// ascending — a fixed association, bitwise-reproducible run to run.
+__global__ void biasrelu_db_chunk_kernel(const float* __restrict__ partials,
+                                         float* __restrict__ chunk_out, int C,
+                                         int nblocks, int chunk) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const int b0 = blockIdx.y * chunk;
+  const int b1 = min(b0 + chunk, nblocks);
+  float s = 0.f;
+  for (int b = b0; b < b1; ++b) s += partials[(long long)b * C + c];
+  chunk_out[(long long)blockIdx.y * C + c] = s;
+}
+
 __global__ void biasrelu_db_finalize_kernel(const float* __restrict__ partials,
                                             float* __restrict__ db, int C,
                                             int nblocks) {
@@ -179,6 +203,25 @@ __global__ void biasrelu_db_finalize_kernel(const float* __restrict__ partials,
     for (int b = 0; b < nblocks; ++b) s += partials[(long long)b * C + c];
     db[c] = s;
   }
+}
+
+// two-stage deterministic column sum of partials[nblocks][C] -> db[C]
+static void br_db_reduce(const torch::Tensor& partials, torch::Tensor& db,
+                         int C, int nblocks, hipStream_t stream) {
+  constexpr int NCHUNK = 16;
+  if (nblocks <= 64) {
+    biasrelu_db_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+        partials.data_ptr<float>(), db.data_ptr<float>(), C, nblocks);
+    return;
+  }
+  const int chunk = (nblocks + NCHUNK - 1) / NCHUNK;
+  const int ny = (nblocks + chunk - 1) / chunk;
+  auto chunks = torch::empty({ny, C}, partials.options());
+  dim3 grid((C + 255) / 256, ny);
+  biasrelu_db_chunk_kernel<<<grid, 256, 0, stream>>>(
+      partials.data_ptr<float>(), chunks.data_ptr<float>(), C, nblocks, chunk);
+  biasrelu_db_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
+      chunks.data_ptr<float>(), db.data_ptr<float>(), C, ny);
 }
 
 // ---------------------------------------------------------------------------
@@ -191,10 +234,10 @@ static bool br_nhwc(const torch::Tensor& t) {
 
 // grid size such that (grid * NPAIR_BLOCK) % q == 0 (q = C/VEC): pins every
 // thread's channel window across grid-stride iterations
-static int br_fixed_grid(long long groups, int q) {
+static int br_fixed_grid(long long groups, int q, int maxgrid) {
   long long desired = (groups + NPAIR_BLOCK * 4 - 1) / (NPAIR_BLOCK * 4);
   if (desired < 256) desired = 256;
-  if (desired > 4096) desired = 4096;
+  if (desired > maxgrid) desired = maxgrid;
   const int m = q / std::gcd((long long)q, (long long)NPAIR_BLOCK);
   return (int)((desired + m - 1) / m * m);
 }
@@ -217,7 +260,7 @@ torch::Tensor biasrelu_fwd(torch::Tensor x, torch::Tensor bias) {
     constexpr int V = Pack16<T>::N;
     if (nhwc && C % V == 0) {
       const long long groups = total / V;
-      const int grid = br_fixed_grid(groups, (int)(C / V));
+      const int grid = br_fixed_grid(groups, (int)(C / V), 4096);
       biasrelu_fwd_vec_kernel<T><<<grid, NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(xc.data_ptr()), bias.data_ptr<float>(),
           reinterpret_cast<T*>(y.data_ptr()), groups, (int)C);
@@ -250,9 +293,11 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
     TORCH_CHECK((std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>),
                 "biasrelu: bf16/fp32 only");
     constexpr int V = Pack16<T>::N;
-    if (nhwc && C % V == 0) {
+    if (nhwc && C % V == 0 && C / V <= NPAIR_BLOCK) {
       const long long groups = total / V;
-      const int grid = br_fixed_grid(groups, (int)(C / V));
+      // smaller grid than the forward: the cross-block finalize reads
+      // grid*C partials serially per channel, so fewer/larger blocks win
+      const int grid = br_fixed_grid(groups, (int)(C / V), 1280);
       auto partials = torch::empty({grid, C}, y.options().dtype(torch::kFloat32));
       biasrelu_bwd_vec_kernel<T><<<grid, NPAIR_BLOCK,
                                    (size_t)NPAIR_BLOCK * V * sizeof(float), stream>>>(
@@ -260,8 +305,7 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
           reinterpret_cast<const T*>(dyc.data_ptr()),
           reinterpret_cast<T*>(dx.data_ptr()), partials.data_ptr<float>(),
           groups, (int)C);
-      biasrelu_db_finalize_kernel<<<(int)std::min<long long>((C + 255) / 256, 64), 256, 0, stream>>>(
-          partials.data_ptr<float>(), db.data_ptr<float>(), (int)C, grid);
+      br_db_reduce(partials, db, (int)C, grid, stream);
     } else {
       const int blocks = (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 1024);
       auto partials = torch::empty({blocks, C}, y.options().dtype(torch::kFloat32));
@@ -270,8 +314,7 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
           reinterpret_cast<const T*>(dyc.data_ptr()),
           reinterpret_cast<T*>(dx.data_ptr()), partials.data_ptr<float>(),
           total, (int)C, cstride);
-      biasrelu_db_finalize_kernel<<<(int)std::min<long long>((C + 255) / 256, 64), 256, 0, stream>>>(
-          partials.data_ptr<float>(), db.data_ptr<float>(), (int)C, blocks);
+      br_db_reduce(partials, db, (int)C, blocks, stream);
     }
   });
   HIP_CHECK_LAST();
