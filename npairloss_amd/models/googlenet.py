@@ -17,7 +17,8 @@ from typing import Dict, Tuple
 import torch
 from torch import nn
 
-from ..ops.vision import ConvBiasReLU, CrossChannelLRN, MaxPool3x3
+from ..ops.vision import (Conv1x1BiasReLU, ConvBiasReLU, CrossChannelLRN,
+                          MaxPool3x3)
 
 
 class ConvReLU(nn.Module):
@@ -31,10 +32,14 @@ class ConvReLU(nn.Module):
 
 
 def _conv_block(cin, cout, k, stride=1, pad=0, fused=True):
-    """conv+bias+relu unit.  fused=True (default) keeps bias OUT of the conv
-    and runs the fused BiasReLU kernel: backward collapses torch's
-    threshold_backward + generic bias-grad reduce into one pass
-    (csrc/biasrelu.hip); fused=False is the plain conv(bias)+ReLU pair."""
+    """conv+bias+relu unit.  fused=True (default): 1x1 convs become ONE
+    fused MFMA GEMM + bias + ReLU pass (csrc/conv1x1.hip); other kernels
+    keep bias OUT of the conv and run the fused BiasReLU kernel, whose
+    backward collapses torch's threshold_backward + generic bias-grad
+    reduce into one pass (csrc/biasrelu.hip).  fused=False is the plain
+    conv(bias)+ReLU pair."""
+    if fused and k == 1 and stride == 1 and pad == 0:
+        return Conv1x1BiasReLU(cin, cout)
     if fused:
         return ConvBiasReLU(cin, cout, k, stride=stride, pad=pad)
     return ConvReLU(cin, cout, k, stride=stride, pad=pad)
@@ -108,7 +113,7 @@ class GoogLeNet(nn.Module):
                 nn.init.xavier_uniform_(m.weight)
                 if m.bias is not None:
                     nn.init.constant_(m.bias, 0.2)  # def.prototxt:109-112 filler
-            elif isinstance(m, ConvBiasReLU):
+            elif isinstance(m, (ConvBiasReLU, Conv1x1BiasReLU)):
                 nn.init.constant_(m.bias, 0.2)  # same filler, bias lives outside conv
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

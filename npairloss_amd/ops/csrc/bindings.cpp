@@ -44,9 +44,13 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
 torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
                            int64_t H, int64_t W);
 
-// biasrelu.hip (EXPERIMENTAL — see the file header)
+// biasrelu.hip
 torch::Tensor biasrelu_fwd(torch::Tensor x, torch::Tensor bias);
 std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy);
+
+// conv1x1.hip
+torch::Tensor conv1x1_bias_relu_fwd(torch::Tensor x, torch::Tensor w, torch::Tensor bias);
+torch::Tensor conv1x1_dgrad(torch::Tensor g, torch::Tensor wt);
 
 // gemm_lowp.hip
 torch::Tensor sim_gemm_nt_bf16(torch::Tensor A, torch::Tensor B);
@@ -76,8 +80,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sim_gemm_nt_bf16", &sim_gemm_nt_bf16, "bf16 MFMA similarity GEMM");
   m.def("sim_gemm_nt_fp8", &sim_gemm_nt_fp8, "fp8 e4m3 MFMA similarity GEMM");
   m.def("cast_fp8", &cast_fp8, "fp32 -> fp8 e4m3 bit pattern");
-  m.def("biasrelu_fwd", &biasrelu_fwd, "fused bias+relu forward (experimental)");
-  m.def("biasrelu_bwd", &biasrelu_bwd, "fused drelu+bias-grad backward (experimental)");
+  m.def("biasrelu_fwd", &biasrelu_fwd, "fused bias+relu forward");
+  m.def("biasrelu_bwd", &biasrelu_bwd, "fused drelu+bias-grad backward");
+  m.def("conv1x1_bias_relu_fwd", &conv1x1_bias_relu_fwd,
+        "fused 1x1-conv GEMM + bias + relu (bf16 MFMA)");
+  m.def("conv1x1_dgrad", &conv1x1_dgrad, "1x1-conv data gradient GEMM (bf16 MFMA)");
   m.def("gemm_nn", &gemm_nn, "fp32 MFMA GEMM A @ B");
   m.def("gemm_tn", &gemm_tn, "fp32 MFMA GEMM A^T @ B");
 }

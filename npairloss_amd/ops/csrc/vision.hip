@@ -490,6 +490,92 @@ __global__ void maxpool3_bwd_kernel(const T* __restrict__ dy,
 }
 
 // ---------------------------------------------------------------------------
+// stride-1 rolling-column backward (NHWC, V=4): one thread walks a whole
+// input row (b, h, cv).  The <=9 covering windows of consecutive w overlap
+// in 6 columns, so instead of re-gathering 9 (idx, dy) pairs per element
+// (the generic kernel above), a 3-column register ring shifts left each
+// step and loads only the NEW column (3 rows) — 1/3 the gather traffic.
+// Fully deterministic (pure gather, fixed order).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+struct MpCol {
+  float g[3][4];       // dy values for the 3 covering rows x 4 channels
+  unsigned int id[3];  // packed uchar4 argmax codes per row (0xFF = invalid)
+};
+
+template <typename T>
+DEVINL void mp_load_col(MpCol<T>& c, const T* __restrict__ dy,
+                        const unsigned char* __restrict__ idx, int ow, int OW,
+                        long long ysw, const long long yrow[3], const int rvalid[3]) {
+#pragma unroll
+  for (int rr = 0; rr < 3; ++rr) {
+    if (rvalid[rr] && ow >= 0 && ow < OW) {
+      const long long a = yrow[rr] + (long long)ow * ysw;
+      c.id[rr] = *reinterpret_cast<const unsigned int*>(idx + a);
+      ld4(dy + a, c.g[rr]);
+    } else {
+      c.id[rr] = 0xFFFFFFFFu;
+      c.g[rr][0] = c.g[rr][1] = c.g[rr][2] = c.g[rr][3] = 0.f;
+    }
+  }
+}
+
+template <typename T>
+__global__ void maxpool3_bwd_s1_row_kernel(const T* __restrict__ dy,
+                                           const unsigned char* __restrict__ idx,
+                                           T* __restrict__ dx,
+                                           int B, int Cv, int H, int W,
+                                           int OH, int OW,
+                                           long long xsb, long long xsh,
+                                           long long ysb, long long ysh) {
+  // NHWC: xsw == ysw == C (channel stride 1); addresses below bake in V=4.
+  const unsigned int total = (unsigned int)((long long)B * H * Cv);
+  for (unsigned int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    const int cv = i % (unsigned int)Cv;
+    const unsigned int r = i / (unsigned int)Cv;
+    const int h = r % (unsigned int)H;
+    const int b = r / (unsigned int)H;
+    // covering output rows oh = h-1+rr, rr in 0..2; dh = h-oh+1 = 2-rr
+    long long yrow[3];
+    int rvalid[3];
+#pragma unroll
+    for (int rr = 0; rr < 3; ++rr) {
+      const int oh = h - 1 + rr;
+      rvalid[rr] = (oh >= 0 && oh < OH);
+      yrow[rr] = (long long)b * ysb + (long long)oh * ysh + (long long)(cv * 4);
+    }
+    const long long ysw = (long long)Cv * 4;  // NHWC w-stride = C
+    MpCol<T> colA, colB, colC;  // ow = w-1, w, w+1
+    mp_load_col(colA, dy, idx, -1, OW, ysw, yrow, rvalid);
+    mp_load_col(colB, dy, idx, 0, OW, ysw, yrow, rvalid);
+    long long xi = (long long)b * xsb + (long long)h * xsh + (long long)(cv * 4);
+    for (int w = 0; w < W; ++w, xi += (long long)Cv * 4) {
+      mp_load_col(colC, dy, idx, w + 1, OW, ysw, yrow, rvalid);
+      float acc[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int rr = 0; rr < 3; ++rr) {
+        // want codes: (dh=2-rr)*3 + dw, dw = 2 (colA), 1 (colB), 0 (colC)
+        const unsigned int wantA = (2 - rr) * 3 + 2;
+        const unsigned int wantB = (2 - rr) * 3 + 1;
+        const unsigned int wantC = (2 - rr) * 3 + 0;
+#pragma unroll
+        for (int v = 0; v < 4; ++v) {
+          const int sh = 8 * v;
+          if (((colA.id[rr] >> sh) & 0xFFu) == wantA) acc[v] += colA.g[rr][v];
+          if (((colB.id[rr] >> sh) & 0xFFu) == wantB) acc[v] += colB.g[rr][v];
+          if (((colC.id[rr] >> sh) & 0xFFu) == wantC) acc[v] += colC.g[rr][v];
+        }
+      }
+      st4(dx + xi, acc);
+      colA = colB;
+      colB = colC;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // host launchers
 // ---------------------------------------------------------------------------
 
@@ -728,7 +814,15 @@ torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
   TORCH_CHECK(total < (1LL << 31), "maxpool3: tensor exceeds the 32-bit index fast path");
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(dy, "maxpool3_bwd", {
-    if (V == 4)
+    if (V == 4 && stride == 1)
+      // rolling-column row walker: 1/3 the gather traffic of the generic
+      // per-element kernel (each thread owns a (b, h, cv) input row)
+      maxpool3_bwd_s1_row_kernel<T><<<grid_for((long long)B * H * (C / 4)),
+                                      NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
+          reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW,
+          xsb, xsh, ysb, ysh);
+    else if (V == 4)
       maxpool3_bwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
           reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW, (int)stride,

@@ -32,6 +32,7 @@ SOURCES = [
     "vision.hip",
     "gemm_lowp.hip",
     "biasrelu.hip",
+    "conv1x1.hip",
 ]
 
 
