@@ -70,12 +70,18 @@ def test_conv1x1_module_end_to_end(amp):
 
     g = torch.randn_like(y2)
     y1.backward(g.to(y1.dtype))
-    y2.backward(g)
-    torch.testing.assert_close(x1.grad.float(), x2.grad, rtol=3e-2, atol=3e-2)
+    # mask-consistent reference backward: near-zero pre-activations can flip
+    # the ReLU mask between bf16 and fp32, giving whole-gradient diffs at
+    # ~3% of sites — use the MODULE's mask so the GEMM chain is what's tested
+    mask = (y1 > 0).float()
+    gm = (g * mask)
+    ref_dx = torch.nn.functional.conv_transpose2d(gm, ref_conv.weight)
+    ref_dw = torch.nn.grad.conv2d_weight(x2.detach(), ref_conv.weight.shape, gm)
+    ref_db = gm.sum(dim=(0, 2, 3))
+    torch.testing.assert_close(x1.grad.float(), ref_dx, rtol=3e-2, atol=3e-2)
     torch.testing.assert_close(m.weight.grad.float().flatten(),
-                               ref_conv.weight.grad.flatten(), rtol=3e-2, atol=3e-2)
-    torch.testing.assert_close(m.bias.grad.float(), ref_conv.bias.grad,
-                               rtol=2e-2, atol=2e-1)
+                               ref_dw.flatten(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(m.bias.grad.float(), ref_db, rtol=2e-2, atol=2e-1)
 
 
 def test_googlenet_fused_close_to_reference_gpu():
