@@ -275,7 +275,10 @@ class Conv1x1BiasReLU(nn.Module):
         self.conv = nn.Conv2d(cin, cout, 1, bias=False)
         self.bias = nn.Parameter(torch.zeros(cout))
         import os
-        self.mode = os.environ.get("NPAIR_CONV1X1", "hybrid")
+        # default "off": measured fastest (profiles/kb_conv1x1 — MIOpen wins
+        # both directions at the bench shapes; hipBLASLt's unsplit TN wgrad
+        # loses badly on the M=800k reduction, our MFMA GEMM is mid-tuning)
+        self.mode = os.environ.get("NPAIR_CONV1X1", "off")
 
     @property
     def weight(self):

@@ -74,9 +74,14 @@ def test_conv1x1_module_end_to_end(amp):
     # the ReLU mask between bf16 and fp32, giving whole-gradient diffs at
     # ~3% of sites — use the MODULE's mask so the GEMM chain is what's tested
     mask = (y1 > 0).float()
-    gm = (g * mask)
-    ref_dx = torch.nn.functional.conv_transpose2d(gm, ref_conv.weight)
-    ref_dw = torch.nn.grad.conv2d_weight(x2.detach(), ref_conv.weight.shape, gm)
+    # reference from the SAME bf16-rounded operands the module consumed
+    # (fp32 operands would differ by input rounding, amplified by the
+    # M-site reduction in dw)
+    gm = (g * mask).to(torch.bfloat16).float()
+    xb = x1.detach().to(torch.bfloat16).float()
+    wb = ref_conv.weight.detach().to(torch.bfloat16).float()
+    ref_dx = torch.nn.functional.conv_transpose2d(gm, wb)
+    ref_dw = torch.nn.grad.conv2d_weight(xb, ref_conv.weight.shape, gm)
     ref_db = gm.sum(dim=(0, 2, 3))
     torch.testing.assert_close(x1.grad.float(), ref_dx, rtol=3e-2, atol=3e-2)
     torch.testing.assert_close(m.weight.grad.float().flatten(),

@@ -41,7 +41,7 @@ def main():
            f"{'matmul+br ms':>13} {'dgrad ms':>9} {'blaslt ms':>10} "
            f"{'mio_bwd':>10} {'blt_bwd':>10}")
     print(hdr)
-    tot = [0.0] * 5
+    tot = [0.0] * 7
     for K, H, N in shapes:
         M = B * H * H
         x = torch.randn(B, K, H, H, device="cuda", dtype=torch.bfloat16)
