@@ -44,8 +44,9 @@ def parse_args():
     p.add_argument("--mining", choices=["production", "hard", "all"], default="production",
                    help="production = GLOBAL RELATIVE_HARD ap + LOCAL HARD an (def.prototxt); "
                         "hard = semi-hard negatives only; all = no mining (RAND)")
-    p.add_argument("--sim-dtype", choices=["fp32", "bf16", "fp8"], default="fp32",
-                   help="similarity-GEMM MFMA precision on GPU")
+    p.add_argument("--sim-dtype", choices=["fp32", "bf16", "fp8", "fp64"], default="fp32",
+                   help="similarity-GEMM precision on GPU (MFMA for "
+                        "fp32/bf16/fp8; rocBLAS DGEMM + templated fp64 kernels)")
     p.add_argument("--timers", action="store_true",
                    help="per-phase HIP-event timing report on stderr (rank 0)")
     p.add_argument("--graph", dest="graph", action="store_true", default=None,

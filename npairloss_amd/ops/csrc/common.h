@@ -77,6 +77,17 @@ struct OpAddD { DEVINL double operator()(double a, double b) const { return a + 
 struct OpAddI { DEVINL int operator()(int a, int b) const { return a + b; } };
 struct OpMinF { DEVINL float operator()(float a, float b) const { return fminf(a, b); } };
 struct OpMaxF { DEVINL float operator()(float a, float b) const { return fmaxf(a, b); } };
+// dtype-generic variants (fp64 path — reference Dtype dispatch, .cu:31-42)
+struct OpMinT { template <typename T> DEVINL T operator()(T a, T b) const { return a < b ? a : b; } };
+struct OpMaxT { template <typename T> DEVINL T operator()(T a, T b) const { return a > b ? a : b; } };
+
+template <typename T> struct DtMax;
+template <> struct DtMax<float> { static constexpr float v = FLT_MAX; };
+template <> struct DtMax<double> { static constexpr double v = DBL_MAX; };
+
+// exp in the storage dtype: fast hardware exp for fp32, libm for fp64
+DEVINL float kexp(float x) { return __expf(x); }
+DEVINL double kexp(double x) { return exp(x); }
 
 // Pair classification: self pair (rank*B + i == j) belongs to NEITHER set
 // (reference GetLabelDiffMtx, .cu:54).
@@ -94,8 +105,10 @@ DEVINL float key_to_float(uint32_t k) {
 }
 
 // Pair-selection rule (reference GetSampledPairMtx, .cu:69-122).
-// thr already includes the margin.
-DEVINL bool select_pos(float s, float thr, int method) {
+// thr already includes the margin.  Templated over float/double (the
+// reference's Dtype dispatch).
+template <typename T>
+DEVINL bool select_pos(T s, T thr, int method) {
   switch (method) {
     case M_HARD: return s < thr;
     case M_EASY: return s >= thr;
@@ -105,7 +118,8 @@ DEVINL bool select_pos(float s, float thr, int method) {
   }
 }
 
-DEVINL bool select_neg(float s, float thr, int method) {
+template <typename T>
+DEVINL bool select_neg(T s, T thr, int method) {
   switch (method) {
     case M_HARD: return s > thr;
     case M_EASY: return s <= thr;

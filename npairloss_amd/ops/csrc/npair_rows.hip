@@ -30,33 +30,34 @@
 // rowstats
 // ---------------------------------------------------------------------------
 
-__global__ void rowstats_kernel(const float* __restrict__ S,
+template <typename T>
+__global__ void rowstats_kernel(const T* __restrict__ S,
                                 const int* __restrict__ lab_l,
                                 const int* __restrict__ lab_g,
                                 int B, int G, int rank,
-                                float* __restrict__ min_within,
-                                float* __restrict__ max_between,
-                                float* __restrict__ max_all) {
-  __shared__ float scratch[NPAIR_BLOCK / WAVE];
+                                T* __restrict__ min_within,
+                                T* __restrict__ max_between,
+                                T* __restrict__ max_all) {
+  __shared__ T scratch[NPAIR_BLOCK / WAVE];
   const int i = blockIdx.x;
   if (i >= B) return;
   const int li = lab_l[i];
-  const float* row = S + (size_t)i * G;
-  float mnw = FLT_MAX, mxb = -FLT_MAX, mxa = -FLT_MAX;
+  const T* row = S + (size_t)i * G;
+  T mnw = DtMax<T>::v, mxb = -DtMax<T>::v, mxa = -DtMax<T>::v;
   for (int j = threadIdx.x; j < G; j += blockDim.x) {
     if (pair_is_self(i, j, rank, B)) continue;
-    const float s = row[j];
+    const T s = row[j];
     if (lab_g[j] == li) {
-      mnw = fminf(mnw, s);
-      mxa = fmaxf(mxa, s);
+      mnw = mnw < s ? mnw : s;
+      mxa = mxa > s ? mxa : s;
     } else {
-      mxb = fmaxf(mxb, s);
-      mxa = fmaxf(mxa, s);
+      mxb = mxb > s ? mxb : s;
+      mxa = mxa > s ? mxa : s;
     }
   }
-  mnw = block_reduce(mnw, OpMinF(), FLT_MAX, scratch);
-  mxb = block_reduce(mxb, OpMaxF(), -FLT_MAX, scratch);
-  mxa = block_reduce(mxa, OpMaxF(), -FLT_MAX, scratch);
+  mnw = block_reduce(mnw, OpMinT(), DtMax<T>::v, scratch);
+  mxb = block_reduce(mxb, OpMaxT(), -DtMax<T>::v, scratch);
+  mxa = block_reduce(mxa, OpMaxT(), -DtMax<T>::v, scratch);
   if (threadIdx.x == 0) {
     min_within[i] = mnw;
     max_between[i] = mxb;
@@ -68,35 +69,36 @@ __global__ void rowstats_kernel(const float* __restrict__ S,
 // fused forward
 // ---------------------------------------------------------------------------
 
-__global__ void fused_fwd_kernel(const float* __restrict__ S,
+template <typename T>
+__global__ void fused_fwd_kernel(const T* __restrict__ S,
                                  const int* __restrict__ lab_l,
                                  const int* __restrict__ lab_g,
                                  int B, int G, int rank,
-                                 const float* __restrict__ thr_p,
-                                 const float* __restrict__ thr_n,
-                                 const float* __restrict__ max_all,
-                                 float margin_ident, float margin_diff,
+                                 const T* __restrict__ thr_p,
+                                 const T* __restrict__ thr_n,
+                                 const T* __restrict__ max_all,
+                                 T margin_ident, T margin_diff,
                                  int ap_method, int an_method,
-                                 float* __restrict__ ident_num,
-                                 float* __restrict__ diff_num,
-                                 float* __restrict__ loss_ident,
-                                 float* __restrict__ loss_sum,
-                                 float* __restrict__ log_term) {
+                                 T* __restrict__ ident_num,
+                                 T* __restrict__ diff_num,
+                                 T* __restrict__ loss_ident,
+                                 T* __restrict__ loss_sum,
+                                 T* __restrict__ log_term) {
   __shared__ double scratch_d[NPAIR_BLOCK / WAVE];
   __shared__ int scratch_i[NPAIR_BLOCK / WAVE];
   const int i = blockIdx.x;
   if (i >= B) return;
   const int li = lab_l[i];
-  const float tp = thr_p[i] + margin_ident;
-  const float tn = thr_n[i] + margin_diff;
-  const float mx = max_all[i];
-  const float* row = S + (size_t)i * G;
+  const T tp = thr_p[i] + margin_ident;
+  const T tn = thr_n[i] + margin_diff;
+  const T mx = max_all[i];
+  const T* row = S + (size_t)i * G;
   int cnt_p = 0, cnt_n = 0;
   double sum_p = 0.0, sum_n = 0.0;
   for (int j = threadIdx.x; j < G; j += blockDim.x) {
     if (pair_is_self(i, j, rank, B)) continue;
-    const float s = row[j];
-    const float e = __expf(s - mx);
+    const T s = row[j];
+    const T e = kexp(s - mx);
     if (lab_g[j] == li) {
       if (select_pos(s, tp, ap_method)) {
         ++cnt_p;
@@ -114,13 +116,13 @@ __global__ void fused_fwd_kernel(const float* __restrict__ S,
   sum_p = block_reduce(sum_p, OpAddD(), 0.0, scratch_d);
   sum_n = block_reduce(sum_n, OpAddD(), 0.0, scratch_d);
   if (threadIdx.x == 0) {
-    ident_num[i] = (float)cnt_p;
-    diff_num[i] = (float)cnt_n;
+    ident_num[i] = (T)cnt_p;
+    diff_num[i] = (T)cnt_n;
     const double lsum = sum_p + sum_n;
-    loss_ident[i] = (float)sum_p;
-    loss_sum[i] = (float)lsum;
+    loss_ident[i] = (T)sum_p;
+    loss_sum[i] = (T)lsum;
     // div + log with the reference's zero-guards (.cu:162-169)
-    log_term[i] = (sum_p == 0.0 || lsum == 0.0) ? 0.f : (float)log(sum_p / lsum);
+    log_term[i] = (sum_p == 0.0 || lsum == 0.0) ? (T)0 : (T)log(sum_p / lsum);
   }
 }
 
@@ -128,40 +130,41 @@ __global__ void fused_fwd_kernel(const float* __restrict__ S,
 // backward weights
 // ---------------------------------------------------------------------------
 
-__global__ void bwd_weights_kernel(const float* __restrict__ S,
+template <typename T>
+__global__ void bwd_weights_kernel(const T* __restrict__ S,
                                    const int* __restrict__ lab_l,
                                    const int* __restrict__ lab_g,
                                    int B, int G, int rank,
-                                   const float* __restrict__ thr_p,
-                                   const float* __restrict__ thr_n,
-                                   const float* __restrict__ max_all,
-                                   const float* __restrict__ loss_ident,
-                                   const float* __restrict__ loss_sum,
-                                   float margin_ident, float margin_diff,
+                                   const T* __restrict__ thr_p,
+                                   const T* __restrict__ thr_n,
+                                   const T* __restrict__ max_all,
+                                   const T* __restrict__ loss_ident,
+                                   const T* __restrict__ loss_sum,
+                                   T margin_ident, T margin_diff,
                                    int ap_method, int an_method,
-                                   float scale,
-                                   float* __restrict__ W) {
+                                   T scale,
+                                   T* __restrict__ W) {
   const size_t total = (size_t)B * G;
   for (size_t idx = (size_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (size_t)gridDim.x * blockDim.x) {
     const int i = idx / G;
     const int j = idx % G;
-    float w = 0.f;
+    T w = (T)0;
     if (!pair_is_self(i, j, rank, B)) {
-      const float s = S[idx];
-      const float e = __expf(s - max_all[i]);
-      const float li = loss_ident[i];
-      const float ls = loss_sum[i];
+      const T s = S[idx];
+      const T e = kexp(s - max_all[i]);
+      const T li = loss_ident[i];
+      const T ls = loss_sum[i];
       if (lab_g[j] == lab_l[i]) {
-        if (select_pos(s, thr_p[i] + margin_ident, ap_method)) {
+        if (select_pos(s, (T)(thr_p[i] + margin_ident), ap_method)) {
           // -part1 + part2 (each guarded on its OWN denominator, .cu:412-417)
-          const float p1 = (li == 0.f) ? 0.f : e / li;
-          const float p2 = (ls == 0.f) ? 0.f : e / ls;
+          const T p1 = (li == (T)0) ? (T)0 : e / li;
+          const T p2 = (ls == (T)0) ? (T)0 : e / ls;
           w = -p1 + p2;
         }
       } else {
-        if (select_neg(s, thr_n[i] + margin_diff, an_method)) {
-          w = (ls == 0.f) ? 0.f : e / ls;  // part3
+        if (select_neg(s, (T)(thr_n[i] + margin_diff), an_method)) {
+          w = (ls == (T)0) ? (T)0 : e / ls;  // part3
         }
       }
     }
@@ -175,29 +178,30 @@ __global__ void bwd_weights_kernel(const float* __restrict__ S,
 
 #define RECALL_MAX_TOPK 15  // reference top list {1,5,10,15}, .cu:390-394
 
-__global__ void recall_kernel(const float* __restrict__ S,
+template <typename T>
+__global__ void recall_kernel(const T* __restrict__ S,
                               const int* __restrict__ lab_l,
                               const int* __restrict__ lab_g,
                               int B, int G, int rank,
                               const int* __restrict__ ks, int nk, int kmax,
                               int* __restrict__ hits) {
   // one block per query row; M = min(kmax, G-2) + 1 threshold candidates
-  __shared__ float cand[NPAIR_BLOCK];
-  __shared__ float extracted[RECALL_MAX_TOPK + 1];
+  __shared__ T cand[NPAIR_BLOCK];
+  __shared__ T extracted[RECALL_MAX_TOPK + 1];
   __shared__ int scratch_i[NPAIR_BLOCK / WAVE];
   const int i = blockIdx.x;
   if (i >= B) return;
   const int li = lab_l[i];
-  const float* row = S + (size_t)i * G;
+  const T* row = S + (size_t)i * G;
 
   // phase 1: per-thread sorted (desc) local top-M list in registers
-  float loc[RECALL_MAX_TOPK + 1];
+  T loc[RECALL_MAX_TOPK + 1];
   const int M = min(kmax, G - 2) + 1;  // threshold index is min(k, len-1), len = G-1
 #pragma unroll
-  for (int m = 0; m <= RECALL_MAX_TOPK; ++m) loc[m] = -FLT_MAX;
+  for (int m = 0; m <= RECALL_MAX_TOPK; ++m) loc[m] = -DtMax<T>::v;
   for (int j = threadIdx.x; j < G; j += blockDim.x) {
     if (pair_is_self(i, j, rank, B)) continue;
-    float s = row[j];
+    T s = row[j];
     if (s <= loc[M - 1]) continue;
     // insertion into the M-length sorted list
     int p = M - 1;
@@ -211,25 +215,27 @@ __global__ void recall_kernel(const float* __restrict__ S,
   // phase 2: cooperative extraction of the global top-M of the row
   int head = 0;
   for (int m = 0; m < M; ++m) {
-    cand[threadIdx.x] = (head < M) ? loc[head] : -FLT_MAX;
+    cand[threadIdx.x] = (head < M) ? loc[head] : -DtMax<T>::v;
     __syncthreads();
     // tree argmax over 256 candidates
     for (int stride = NPAIR_BLOCK / 2; stride > 0; stride >>= 1) {
       if (threadIdx.x < stride)
-        cand[threadIdx.x] = fmaxf(cand[threadIdx.x], cand[threadIdx.x + stride]);
+        cand[threadIdx.x] = cand[threadIdx.x] > cand[threadIdx.x + stride]
+                                ? cand[threadIdx.x] : cand[threadIdx.x + stride];
       __syncthreads();
     }
-    const float winner = cand[0];
+    const T winner = cand[0];
     __syncthreads();
     if (threadIdx.x == 0) extracted[m] = winner;
     // exactly ONE thread (the lowest-id holder) pops its head
     const bool mine = (head < M) && (loc[head] == winner);
     // ballot across block via LDS: find the lowest thread id holding winner
-    cand[threadIdx.x] = mine ? (float)threadIdx.x : (float)NPAIR_BLOCK;
+    cand[threadIdx.x] = mine ? (T)threadIdx.x : (T)NPAIR_BLOCK;
     __syncthreads();
     for (int stride = NPAIR_BLOCK / 2; stride > 0; stride >>= 1) {
       if (threadIdx.x < stride)
-        cand[threadIdx.x] = fminf(cand[threadIdx.x], cand[threadIdx.x + stride]);
+        cand[threadIdx.x] = cand[threadIdx.x] < cand[threadIdx.x + stride]
+                                ? cand[threadIdx.x] : cand[threadIdx.x + stride];
       __syncthreads();
     }
     if ((int)cand[0] == (int)threadIdx.x) ++head;
@@ -241,7 +247,7 @@ __global__ void recall_kernel(const float* __restrict__ S,
     const int k = ks[ki];
     const int ti = min(k, G - 2);
     if (ti < 0) continue;  // G < 2: no retrievable database
-    const float thr = extracted[min(ti, M - 1)];
+    const T thr = extracted[min(ti, M - 1)];
     int hit = 0;
     for (int j = threadIdx.x; j < G; j += blockDim.x) {
       if (pair_is_self(i, j, rank, B)) continue;
@@ -259,8 +265,9 @@ __global__ void recall_kernel(const float* __restrict__ S,
 
 static void check_sg(const torch::Tensor& S, const torch::Tensor& lab_l,
                      const torch::Tensor& lab_g) {
-  TORCH_CHECK(S.is_cuda() && S.dtype() == torch::kFloat32 && S.is_contiguous(),
-              "S must be contiguous fp32 on GPU");
+  TORCH_CHECK(S.is_cuda() && S.is_contiguous() &&
+                  (S.dtype() == torch::kFloat32 || S.dtype() == torch::kFloat64),
+              "S must be contiguous fp32/fp64 on GPU");
   TORCH_CHECK(lab_l.dtype() == torch::kInt32 && lab_g.dtype() == torch::kInt32,
               "labels must be int32");
   TORCH_CHECK(S.size(0) == lab_l.numel() && S.size(1) == lab_g.numel(),
@@ -276,9 +283,12 @@ std::vector<torch::Tensor> rowstats(torch::Tensor S, torch::Tensor lab_l,
   auto mxb = torch::empty({B}, opts);
   auto mxa = torch::empty({B}, opts);
   auto stream = at::hip::getCurrentHIPStream();
-  rowstats_kernel<<<B, NPAIR_BLOCK, 0, stream>>>(
-      S.data_ptr<float>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
-      (int)rank, mnw.data_ptr<float>(), mxb.data_ptr<float>(), mxa.data_ptr<float>());
+  AT_DISPATCH_FLOATING_TYPES(S.scalar_type(), "rowstats", [&] {
+    rowstats_kernel<scalar_t><<<B, NPAIR_BLOCK, 0, stream>>>(
+        S.data_ptr<scalar_t>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
+        (int)rank, mnw.data_ptr<scalar_t>(), mxb.data_ptr<scalar_t>(),
+        mxa.data_ptr<scalar_t>());
+  });
   HIP_CHECK_LAST();
   return {mnw, mxb, mxa};
 }
@@ -298,13 +308,15 @@ std::vector<torch::Tensor> fused_fwd(torch::Tensor S, torch::Tensor lab_l,
   auto loss_sum = torch::empty({B}, opts);
   auto log_term = torch::empty({B}, opts);
   auto stream = at::hip::getCurrentHIPStream();
-  fused_fwd_kernel<<<B, NPAIR_BLOCK, 0, stream>>>(
-      S.data_ptr<float>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
-      (int)rank, thr_p.data_ptr<float>(), thr_n.data_ptr<float>(),
-      max_all.data_ptr<float>(), (float)margin_ident, (float)margin_diff,
-      (int)ap_method, (int)an_method, ident_num.data_ptr<float>(),
-      diff_num.data_ptr<float>(), loss_ident.data_ptr<float>(),
-      loss_sum.data_ptr<float>(), log_term.data_ptr<float>());
+  AT_DISPATCH_FLOATING_TYPES(S.scalar_type(), "fused_fwd", [&] {
+    fused_fwd_kernel<scalar_t><<<B, NPAIR_BLOCK, 0, stream>>>(
+        S.data_ptr<scalar_t>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
+        (int)rank, thr_p.data_ptr<scalar_t>(), thr_n.data_ptr<scalar_t>(),
+        max_all.data_ptr<scalar_t>(), (scalar_t)margin_ident, (scalar_t)margin_diff,
+        (int)ap_method, (int)an_method, ident_num.data_ptr<scalar_t>(),
+        diff_num.data_ptr<scalar_t>(), loss_ident.data_ptr<scalar_t>(),
+        loss_sum.data_ptr<scalar_t>(), log_term.data_ptr<scalar_t>());
+  });
   HIP_CHECK_LAST();
   return {ident_num, diff_num, loss_ident, loss_sum, log_term};
 }
@@ -322,12 +334,14 @@ torch::Tensor bwd_weights(torch::Tensor S, torch::Tensor lab_l,
   const size_t total = (size_t)B * G;
   const int blocks = (int)min((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, (size_t)2048);
   auto stream = at::hip::getCurrentHIPStream();
-  bwd_weights_kernel<<<blocks, NPAIR_BLOCK, 0, stream>>>(
-      S.data_ptr<float>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
-      (int)rank, thr_p.data_ptr<float>(), thr_n.data_ptr<float>(),
-      max_all.data_ptr<float>(), loss_ident.data_ptr<float>(),
-      loss_sum.data_ptr<float>(), (float)margin_ident, (float)margin_diff,
-      (int)ap_method, (int)an_method, (float)scale, W.data_ptr<float>());
+  AT_DISPATCH_FLOATING_TYPES(S.scalar_type(), "bwd_weights", [&] {
+    bwd_weights_kernel<scalar_t><<<blocks, NPAIR_BLOCK, 0, stream>>>(
+        S.data_ptr<scalar_t>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
+        (int)rank, thr_p.data_ptr<scalar_t>(), thr_n.data_ptr<scalar_t>(),
+        max_all.data_ptr<scalar_t>(), loss_ident.data_ptr<scalar_t>(),
+        loss_sum.data_ptr<scalar_t>(), (scalar_t)margin_ident, (scalar_t)margin_diff,
+        (int)ap_method, (int)an_method, (scalar_t)scale, W.data_ptr<scalar_t>());
+  });
   HIP_CHECK_LAST();
   return W;
 }
@@ -343,9 +357,11 @@ torch::Tensor recall_hits(torch::Tensor S, torch::Tensor lab_l,
   TORCH_CHECK(kmax >= 1 && kmax <= RECALL_MAX_TOPK, "k in [1,15]");
   auto hits = torch::zeros({nk}, S.options().dtype(torch::kInt32));
   auto stream = at::hip::getCurrentHIPStream();
-  recall_kernel<<<B, NPAIR_BLOCK, 0, stream>>>(
-      S.data_ptr<float>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
-      (int)rank, ks_t.data_ptr<int>(), nk, (int)kmax, hits.data_ptr<int>());
+  AT_DISPATCH_FLOATING_TYPES(S.scalar_type(), "recall_hits", [&] {
+    recall_kernel<scalar_t><<<B, NPAIR_BLOCK, 0, stream>>>(
+        S.data_ptr<scalar_t>(), lab_l.data_ptr<int>(), lab_g.data_ptr<int>(), B, G,
+        (int)rank, ks_t.data_ptr<int>(), nk, (int)kmax, hits.data_ptr<int>());
+  });
   HIP_CHECK_LAST();
   return hits;
 }

@@ -235,12 +235,16 @@ def _forward_hip(F_l, labels_l, F_g, labels_g, rank, cfg: NPairLossConfig, ks,
                  sim_dtype: str = "fp32"):
     C = _backend.ext()
     B = F_l.shape[0]
-    # similarity GEMM precision: fp32 MFMA (exact, default), bf16 MFMA, or
-    # fp8-e4m3 MFMA (unit-norm embeddings fit e4m3 without scaling)
+    # similarity GEMM precision: fp32 MFMA (exact, default), bf16 MFMA,
+    # fp8-e4m3 MFMA (unit-norm embeddings fit e4m3 without scaling), or
+    # fp64 (rocBLAS DGEMM; every row kernel below is float/double templated
+    # like the reference's Dtype dispatch, .cu:31-42)
     if sim_dtype == "bf16":
         S = C.sim_gemm_nt_bf16(F_l.to(torch.bfloat16), F_g.to(torch.bfloat16))
     elif sim_dtype == "fp8":
         S = C.sim_gemm_nt_fp8(C.cast_fp8(F_l), C.cast_fp8(F_g))
+    elif sim_dtype == "fp64":
+        S = (F_l @ F_g.t()).contiguous()  # fp64 operands -> rocBLAS DGEMM
     else:
         S = C.sim_gemm_nt(F_l, F_g)
     lab_l = labels_l.to(torch.int32)
@@ -290,7 +294,8 @@ class _NPairLossFn(torch.autograd.Function):
     def forward(ctx, features: torch.Tensor, labels: torch.Tensor, cfg: NPairLossConfig,
                 ks: Tuple[int, ...], group, sim_dtype: str = "fp32"):
         in_dtype = features.dtype
-        F_l = features.detach().float().contiguous()
+        compute_dtype = torch.float64 if sim_dtype == "fp64" else torch.float32
+        F_l = features.detach().to(compute_dtype).contiguous()
         labels = labels.detach()
         ws = comm.world_size(group)
         rank = comm.rank(group)
@@ -331,8 +336,12 @@ class _NPairLossFn(torch.autograd.Function):
                               cfg.margin_ident, cfg.margin_diff,
                               int(cfg.ap_mining_method), int(cfg.an_mining_method), 1.0 / B)
             W = W * dloss  # scalar broadcast, stays on device
-            dF_l = C.gemm_nn(W, F_g)
-            dF_t = C.gemm_tn(W, F_l)
+            if S.dtype == torch.float64:
+                dF_l = W @ F_g           # rocBLAS DGEMM (fp64 path)
+                dF_t = W.t().contiguous() @ F_l
+            else:
+                dF_l = C.gemm_nn(W, F_g)
+                dF_t = C.gemm_tn(W, F_l)
         else:
             W = _bwd_weights_torch(S, labels_l, labels_g, ctx.rank, thr_p, thr_n, max_all,
                                    loss_ident, loss_sum, cfg, 1.0 / B) * dloss
@@ -369,7 +378,7 @@ class NPairMultiClassLoss(nn.Module):
                  top_k: Tuple[int, ...] = (1, 5, 10), group=None,
                  sim_dtype: str = "fp32"):
         super().__init__()
-        assert sim_dtype in ("fp32", "bf16", "fp8")
+        assert sim_dtype in ("fp32", "bf16", "fp8", "fp64")
         self.cfg = cfg if cfg is not None else NPairLossConfig()
         self.top_k = tuple(top_k)
         self.group = group
