@@ -73,6 +73,8 @@ def build_trainer_from_prototxt(
     channels_last: Optional[bool] = None,
     pure_bf16: bool = False,
     backbone: Optional[str] = None,
+    hip_graph: Optional[bool] = None,
+    timers: bool = False,
 ) -> Trainer:
     net = parse_net_prototxt(_clean_prototxt(net_text))
 
@@ -161,7 +163,8 @@ def build_trainer_from_prototxt(
         channels_last = torch.cuda.is_available()
     trainer = Trainer(model, loss_mod, solver, loader, test_loader=test_loader,
                       device=device, amp_dtype=amp_dtype,
-                      channels_last=channels_last, pure_bf16=pure_bf16)
+                      channels_last=channels_last, pure_bf16=pure_bf16,
+                      hip_graph=hip_graph, timers=timers)
     trainer.augment = aug  # applied in train_step before the model (TRAIN phase)
     # Caffe transform_param (mean subtraction / crop) applies to real 0-255
     # image sources; synthetic data is already zero-mean at crop size.

@@ -46,9 +46,19 @@ class Trainer:
                  amp_dtype: Optional[torch.dtype] = None,
                  log_fn: Callable[[str], None] = print,
                  channels_last: bool = False,
-                 pure_bf16: bool = False):
+                 pure_bf16: bool = False,
+                 hip_graph: Optional[bool] = None,
+                 timers: bool = False):
         self.device = device or (torch.device("cuda") if torch.cuda.is_available()
                                  else torch.device("cpu"))
+        # hipGraph whole-step replay (the bench's headline mode): capture
+        # after `graph_warmup` eager steps, replay thereafter; off on CPU.
+        self.use_graph = hip_graph if hip_graph is not None else False
+        self.graph_warmup = 12
+        self._graph_state = None
+        self._graph_warmup_left = self.graph_warmup
+        from ..utils.profiling import PhaseTimers
+        self.timers = PhaseTimers(enabled=timers, use_cuda=self.device.type == "cuda")
         self.model = model.to(self.device)
         if channels_last:
             self.model = self.model.to(memory_format=torch.channels_last)
@@ -102,9 +112,10 @@ class Trainer:
 
     # -- one training iteration --------------------------------------------
 
-    def train_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
-        self.model.train()
-        images = images.to(self.device, non_blocking=True)
+    def _compute_step(self, images: torch.Tensor, labels: torch.Tensor):
+        """Device-only step body (no host reads): preprocess + augment +
+        forward + loss + backward + grad comm + optimizer.  hipGraph-
+        capturable — inputs must sit at fixed addresses during capture."""
         if self.preprocess is not None and images.dim() == 4:
             from ..data.transforms import preprocess as _pp
             images = _pp(images, self.preprocess)
@@ -112,11 +123,8 @@ class Trainer:
             images = self.augment(images)
         if self.channels_last and images.dim() == 4:
             images = images.to(memory_format=torch.channels_last)
-        labels = labels.to(self.device, non_blocking=True)
         if self.pure_bf16:
             images = images.to(torch.bfloat16)
-        lr = self.solver.lr_at(self.iter)
-        self.optimizer.set_lr(lr)
         self.reducer.zero_grad()  # zeroes the persistent flat grad buckets
         if self.amp_dtype is not None and self.device.type == "cuda":
             with torch.autocast("cuda", dtype=self.amp_dtype):
@@ -128,12 +136,85 @@ class Trainer:
         out.loss.backward()
         self.reducer.finalize()
         self.optimizer.step()
+        return out
+
+    def train_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
+        self.model.train()
+        images = images.to(self.device, non_blocking=True)
+        labels = labels.to(self.device, non_blocking=True)
+        lr = self.solver.lr_at(self.iter)
+        self.optimizer.set_lr(lr)
+        with self.timers.phase("step"):
+            out = self._compute_step(images, labels)
         self.iter += 1
         if self.divergence_check and self.iter % self.divergence_check == 0:
             lv = float(out.loss.detach())
             if not (lv == lv and abs(lv) != float("inf")):
                 raise FloatingPointError(
                     f"training diverged: loss={lv} at iter {self.iter}")
+        return {
+            "loss": out.loss.detach(), "top1": out.retrieve_top1,
+            "top5": out.retrieve_top5, "top10": out.retrieve_top10,
+            "asum": out.feature_asum, "lr": lr,
+        }
+
+    # -- hipGraph-captured iteration ----------------------------------------
+
+    def _graph_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
+        """Like train_step, but the whole compute step replays as ONE
+        hipGraph: batches are copied into static device tensors, the
+        captured work (fwd + loss + bwd + RCCL comm + optimizer) re-executes
+        with zero per-kernel launch gaps.  Capture happens lazily after
+        `graph_warmup` eager steps (MIOpen find must have run) and is
+        re-taken whenever the LR-policy value changes (lr is baked into the
+        captured optimizer kernels)."""
+        self.model.train()
+        lr = self.solver.lr_at(self.iter)
+        g = self._graph_state
+        if g is not None and g["lr"] != lr:
+            g = self._graph_state = None  # lr changed: recapture
+        if g is None:
+            if self._graph_warmup_left > 0:
+                self._graph_warmup_left -= 1
+                return self.train_step(images, labels)
+            self.optimizer.set_lr(lr)
+            static_img = images.to(self.device, non_blocking=True).clone()
+            static_lab = labels.to(self.device, non_blocking=True).clone()
+            torch.cuda.synchronize()
+            if comm.is_dist() and comm.world_size() > 1:
+                import torch.distributed as dist
+                dist.barrier()  # align ranks: every rank captures the same comm
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._compute_step(static_img, static_lab)  # allocator warmup
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            try:
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    out = self._compute_step(static_img, static_lab)
+            except Exception as e:  # noqa: BLE001
+                self.log(f"hipGraph capture failed ({e!r}); continuing eager")
+                self.use_graph = False
+                return self.train_step(images, labels)
+            self._graph_state = g = dict(graph=graph, img=static_img,
+                                         lab=static_lab, out=out, lr=lr)
+            self.iter += 1
+            return self._stats_from(g, lr)
+        g["img"].copy_(images.to(self.device, non_blocking=True))
+        g["lab"].copy_(labels.to(self.device, non_blocking=True))
+        g["graph"].replay()
+        self.iter += 1
+        if self.divergence_check and self.iter % self.divergence_check == 0:
+            lv = float(g["out"].loss.detach())
+            if not (lv == lv and abs(lv) != float("inf")):
+                raise FloatingPointError(
+                    f"training diverged: loss={lv} at iter {self.iter}")
+        return self._stats_from(g, lr)
+
+    def _stats_from(self, g, lr) -> dict:
+        out = g["out"]
         return {
             "loss": out.loss.detach(), "top1": out.retrieve_top1,
             "top5": out.retrieve_top5, "top10": out.retrieve_top10,
@@ -183,7 +264,10 @@ class Trainer:
                     bs.set_epoch(epoch)  # deterministic (base_seed, epoch) reseed
                 data_iter = iter(self.train_loader)
                 images, labels = next(data_iter)
-            stats = self.train_step(images, labels)
+            if self.use_graph and self.device.type == "cuda":
+                stats = self._graph_step(images, labels)
+            else:
+                stats = self.train_step(images, labels)
             if self.solver.display and self.iter % self.solver.display == 0:
                 loss_v = float(stats["loss"])
                 sm = self.avg.add(loss_v)

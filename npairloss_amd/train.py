@@ -29,12 +29,27 @@ def main(argv=None):
     p.add_argument("--num-workers", type=int, default=2)
     p.add_argument("--backbone", default=None,
                    help="override the backbone (googlenet/resnet50/vit)")
+    p.add_argument("--graph", dest="graph", action="store_true", default=None,
+                   help="hipGraph whole-step capture+replay (default ON on "
+                        "GPU; the bench's headline mode — lazy capture after "
+                        "warmup, recapture on LR-policy steps)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
+    p.add_argument("--timers", action="store_true",
+                   help="per-phase HIP-event timing (eager steps only)")
+    p.add_argument("--base-lr", type=float, default=None,
+                   help="override solver base_lr (e.g. warmup demos)")
+    p.add_argument("--display", type=int, default=None,
+                   help="override solver display interval")
     args = p.parse_args(argv)
 
     from .config.params import SolverConfig
     from .engine.net_builder import build_trainer_from_prototxt
 
     solver = SolverConfig.from_prototxt(open(args.solver).read())
+    if args.base_lr is not None:
+        solver.base_lr = args.base_lr
+    if args.display is not None:
+        solver.display = args.display
     net_path = args.net or solver.net
     if net_path and not os.path.isabs(net_path):
         cand = os.path.join(os.path.dirname(os.path.abspath(args.solver)), net_path)
@@ -53,11 +68,12 @@ def main(argv=None):
     amp_dtype = None if args.pure_bf16 else {"off": None, "bf16": torch.bfloat16,
                                               "fp16": torch.float16}[args.amp]
     caffemodel = args.weights if (args.weights and args.weights.endswith(".caffemodel")) else None
+    use_graph = args.graph if args.graph is not None else torch.cuda.is_available()
     trainer = build_trainer_from_prototxt(
         net_text, solver, synthetic_classes=args.synthetic_classes,
         amp_dtype=amp_dtype, caffemodel=caffemodel,
         num_workers=args.num_workers, pure_bf16=args.pure_bf16,
-        backbone=args.backbone)
+        backbone=args.backbone, hip_graph=use_graph, timers=args.timers)
     if args.weights and args.weights.endswith(".pt"):
         trainer.restore(args.weights)
     trainer.fit(max_iter=args.max_iter)
