@@ -112,10 +112,9 @@ class Trainer:
 
     # -- one training iteration --------------------------------------------
 
-    def _compute_step(self, images: torch.Tensor, labels: torch.Tensor):
-        """Device-only step body (no host reads): preprocess + augment +
-        forward + loss + backward + grad comm + optimizer.  hipGraph-
-        capturable — inputs must sit at fixed addresses during capture."""
+    def _prepare(self, images: torch.Tensor) -> torch.Tensor:
+        """Input preprocessing run EAGERLY (outside any captured graph):
+        Caffe transform_param, DataTransformer augmentation, layout/dtype."""
         if self.preprocess is not None and images.dim() == 4:
             from ..data.transforms import preprocess as _pp
             images = _pp(images, self.preprocess)
@@ -125,6 +124,12 @@ class Trainer:
             images = images.to(memory_format=torch.channels_last)
         if self.pure_bf16:
             images = images.to(torch.bfloat16)
+        return images
+
+    def _compute_step(self, images: torch.Tensor, labels: torch.Tensor):
+        """Device-only step body (no host reads): forward + loss + backward
+        + grad comm + optimizer.  hipGraph-capturable — inputs must already
+        be prepared (_prepare) and sit at fixed addresses during capture."""
         self.reducer.zero_grad()  # zeroes the persistent flat grad buckets
         if self.amp_dtype is not None and self.device.type == "cuda":
             with torch.autocast("cuda", dtype=self.amp_dtype):
@@ -144,6 +149,8 @@ class Trainer:
         labels = labels.to(self.device, non_blocking=True)
         lr = self.solver.lr_at(self.iter)
         self.optimizer.set_lr(lr)
+        with self.timers.phase("data_prep"):
+            images = self._prepare(images)
         with self.timers.phase("step"):
             out = self._compute_step(images, labels)
         self.iter += 1
@@ -178,7 +185,7 @@ class Trainer:
                 self._graph_warmup_left -= 1
                 return self.train_step(images, labels)
             self.optimizer.set_lr(lr)
-            static_img = images.to(self.device, non_blocking=True).clone()
+            static_img = self._prepare(images.to(self.device, non_blocking=True)).clone()
             static_lab = labels.to(self.device, non_blocking=True).clone()
             torch.cuda.synchronize()
             if comm.is_dist() and comm.world_size() > 1:
@@ -202,7 +209,7 @@ class Trainer:
                                          lab=static_lab, out=out, lr=lr)
             self.iter += 1
             return self._stats_from(g, lr)
-        g["img"].copy_(images.to(self.device, non_blocking=True))
+        g["img"].copy_(self._prepare(images.to(self.device, non_blocking=True)))
         g["lab"].copy_(labels.to(self.device, non_blocking=True))
         g["graph"].replay()
         self.iter += 1
