@@ -56,7 +56,7 @@ class Trainer:
         self.use_graph = hip_graph if hip_graph is not None else False
         self.graph_warmup = 12
         self._graph_state = None
-        self._graph_warmup_left = self.graph_warmup
+        self._graph_warmup_left = None  # latched from graph_warmup on first step
         from ..utils.profiling import PhaseTimers
         self.timers = PhaseTimers(enabled=timers, use_cuda=self.device.type == "cuda")
         self.model = model.to(self.device)
@@ -180,6 +180,8 @@ class Trainer:
         g = self._graph_state
         if g is not None and g["lr"] != lr:
             g = self._graph_state = None  # lr changed: recapture
+        if self._graph_warmup_left is None:
+            self._graph_warmup_left = self.graph_warmup
         if g is None:
             if self._graph_warmup_left > 0:
                 self._graph_warmup_left -= 1
