@@ -201,7 +201,11 @@ class Trainer:
             torch.cuda.synchronize()
             try:
                 graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
+                # thread_local: only THIS thread's HIP calls are captured;
+                # background threads (DataLoader machinery, autograd engine
+                # workers from earlier eager iterations) otherwise poison
+                # the capture and hipGraphInstantiate crashes
+                with torch.cuda.graph(graph, capture_error_mode="thread_local"):
                     out = self._compute_step(static_img, static_lab)
             except Exception as e:  # noqa: BLE001
                 self.log(f"hipGraph capture failed ({e!r}); continuing eager")
