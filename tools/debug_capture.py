@@ -19,6 +19,10 @@ VARIANTS = {
     4: "trainer capture, synthetic batch, no test_loader, benchmark=True",
     5: "bare capture of trainer._compute_step outside fit, benchmark=True",
     6: "trainer capture, loader batch, workers=0, benchmark=True",
+    7: "manual loop: live loader iterator, SYNTHETIC tensors to _graph_step",
+    8: "manual loop: pre-materialized loader batches (iterator dead)",
+    9: "manual loop: synthetic tensors, stats dict held across calls",
+    10: "manual loop: live iterator batches passed to _graph_step",
 }
 
 
@@ -73,6 +77,37 @@ def run_variant(v: int):
         g.replay()
         torch.cuda.synchronize()
         print("variant 5: bare capture OK")
+        return
+    if v in (7, 9, 10):
+        import itertools
+
+        x = torch.randn(120, 3, 224, 224)
+        lab = torch.arange(60).repeat_interleave(2)
+        it = iter(tr.train_loader) if v in (7, 10) else None
+        stats = None
+        for i in range(10):
+            if v == 10:
+                try:
+                    bx, blab = next(it)
+                except StopIteration:
+                    it = iter(tr.train_loader)
+                    bx, blab = next(it)
+            else:
+                if it is not None:
+                    next(it)  # keep the iterator hot but ignore its batch
+                bx, blab = x, lab
+            s = tr._graph_step(bx, blab)
+            if v == 9:
+                stats = s  # held across the capture call
+        print(f"variant {v}: captured={tr._graph_state is not None} OK")
+        return
+    if v == 8:
+        import itertools
+
+        batches = list(itertools.islice(iter(tr.train_loader), 10))
+        for bx, blab in batches:
+            tr._graph_step(bx, blab)
+        print(f"variant 8: captured={tr._graph_state is not None} OK")
         return
     # 1, 2, 6: run fit for a few iters (captures at graph_warmup)
     tr.fit(max_iter=10)
