@@ -159,10 +159,14 @@ class Trainer:
             if not (lv == lv and abs(lv) != float("inf")):
                 raise FloatingPointError(
                     f"training diverged: loss={lv} at iter {self.iter}")
+        # detach EVERYTHING: a non-detached Function output held across a
+        # later hipGraph capture keeps the autograd node (and its stream
+        # state) alive and hipGraphInstantiate segfaults (bisect: tools/
+        # debug_capture.py — hold-the-stats-dict was the exact trigger)
         return {
-            "loss": out.loss.detach(), "top1": out.retrieve_top1,
-            "top5": out.retrieve_top5, "top10": out.retrieve_top10,
-            "asum": out.feature_asum, "lr": lr,
+            "loss": out.loss.detach(), "top1": out.retrieve_top1.detach(),
+            "top5": out.retrieve_top5.detach(), "top10": out.retrieve_top10.detach(),
+            "asum": out.feature_asum.detach(), "lr": lr,
         }
 
     # -- hipGraph-captured iteration ----------------------------------------
@@ -228,10 +232,14 @@ class Trainer:
 
     def _stats_from(self, g, lr) -> dict:
         out = g["out"]
+        # detach EVERYTHING: a non-detached Function output held across a
+        # later hipGraph capture keeps the autograd node (and its stream
+        # state) alive and hipGraphInstantiate segfaults (bisect: tools/
+        # debug_capture.py — hold-the-stats-dict was the exact trigger)
         return {
-            "loss": out.loss.detach(), "top1": out.retrieve_top1,
-            "top5": out.retrieve_top5, "top10": out.retrieve_top10,
-            "asum": out.feature_asum, "lr": lr,
+            "loss": out.loss.detach(), "top1": out.retrieve_top1.detach(),
+            "top5": out.retrieve_top5.detach(), "top10": out.retrieve_top10.detach(),
+            "asum": out.feature_asum.detach(), "lr": lr,
         }
 
     # -- evaluation ---------------------------------------------------------
