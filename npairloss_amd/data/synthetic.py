@@ -15,7 +15,8 @@ from torch.utils.data import Dataset
 
 class SyntheticImageDataset(Dataset):
     def __init__(self, num_classes: int = 256, per_class: int = 16,
-                 image_size: int = 224, seed: int = 0, noise: float = 0.3):
+                 image_size: int = 224, seed: int = 0, noise: float = 0.3,
+                 cache: bool = None):
         self.num_classes = num_classes
         self.per_class = per_class
         self.image_size = image_size
@@ -25,17 +26,29 @@ class SyntheticImageDataset(Dataset):
         # low-res class patterns, upsampled on access (keeps memory small)
         g = torch.Generator().manual_seed(seed)
         self._patterns = torch.randn(num_classes, 3, 8, 8, generator=g)
+        # generating a 224^2 image (interpolate + randn) costs ~ms on CPU,
+        # which starves the GPU; cache generated images for small datasets
+        # (deterministic per index, so caching changes nothing else)
+        if cache is None:
+            cache = len(self.labels) * image_size * image_size * 12 <= 2 << 30
+        self._cache = {} if cache else None
 
     def __len__(self):
         return len(self.labels)
 
     def __getitem__(self, idx):
+        if self._cache is not None:
+            hit = self._cache.get(idx)
+            if hit is not None:
+                return hit
         lab = self.labels[idx]
         g = torch.Generator().manual_seed(self.seed * 1000003 + idx)
         base = torch.nn.functional.interpolate(
             self._patterns[lab : lab + 1], size=(self.image_size, self.image_size),
             mode="bilinear", align_corners=False)[0]
         img = base + self.noise * torch.randn(3, self.image_size, self.image_size, generator=g)
+        if self._cache is not None:
+            self._cache[idx] = (img, lab)
         return img, lab
 
 

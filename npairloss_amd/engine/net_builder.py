@@ -75,6 +75,7 @@ def build_trainer_from_prototxt(
     backbone: Optional[str] = None,
     hip_graph: Optional[bool] = None,
     timers: bool = False,
+    init: str = "caffe",
 ) -> Trainer:
     net = parse_net_prototxt(_clean_prototxt(net_text))
 
@@ -108,7 +109,7 @@ def build_trainer_from_prototxt(
     # --- backbone: the reference's conv stack is GoogLeNet v1 (overridable);
     # L2-normalize the embedding iff the net has an L2Normalize layer
     has_l2 = bool(net.find("L2Normalize"))
-    model = build_embedding_model(backbone or "googlenet", normalize=has_l2)
+    model = build_embedding_model(backbone or "googlenet", normalize=has_l2, init=init)
     if caffemodel:
         load_caffemodel_into(model, caffemodel)
 

@@ -31,12 +31,13 @@ class EmbeddingNet(nn.Module):
         return f
 
 
-def build_embedding_model(name: str, embed_dim: int = None, normalize: bool = True) -> EmbeddingNet:
+def build_embedding_model(name: str, embed_dim: int = None, normalize: bool = True,
+                          init: str = "caffe") -> EmbeddingNet:
     name = name.lower()
     if name in ("googlenet", "inception_v1"):
         import os
         fused = os.environ.get("NPAIR_FUSED_CONV", "1") != "0"
-        bb = GoogLeNet(fused_bias_relu=fused)
+        bb = GoogLeNet(fused_bias_relu=fused, init=init)
     elif name in ("resnet50", "resnet-50"):
         bb = ResNet50(embed_dim=embed_dim or 128)
     elif name in ("vit", "vit-b/16", "vitb16"):

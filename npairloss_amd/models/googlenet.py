@@ -87,8 +87,17 @@ class GoogLeNet(nn.Module):
 
     embed_dim = 1024
 
-    def __init__(self, dropout: float = 0.4, fused_bias_relu: bool = True):
+    def __init__(self, dropout: float = 0.4, fused_bias_relu: bool = True,
+                 init: str = "caffe"):
+        """init: "caffe" reproduces the reference's filler exactly (xavier
+        weights + constant 0.2 bias, def.prototxt:109-112) — which starts
+        the embedding COLLAPSED (all-positive activations -> all pairwise
+        sims ~1; the reference's 2M-iteration schedule absorbs this).
+        "modern" = kaiming fan-out + zero bias: same architecture, trains
+        from iteration 1 (the documented init deviation for demos)."""
         super().__init__()
+        assert init in ("caffe", "modern")
+        self._init_mode = init
         fused = fused_bias_relu
         self.conv1 = _conv_block(3, 64, 7, stride=2, pad=3, fused=fused)
         self.pool1 = MaxPool3x3(stride=2)
@@ -108,13 +117,19 @@ class GoogLeNet(nn.Module):
         self._init_weights()
 
     def _init_weights(self):
+        caffe = self._init_mode == "caffe"
+        bias_v = 0.2 if caffe else 0.0
         for m in self.modules():
             if isinstance(m, nn.Conv2d):
-                nn.init.xavier_uniform_(m.weight)
+                if caffe:
+                    nn.init.xavier_uniform_(m.weight)
+                else:
+                    nn.init.kaiming_normal_(m.weight, mode="fan_out",
+                                            nonlinearity="relu")
                 if m.bias is not None:
-                    nn.init.constant_(m.bias, 0.2)  # def.prototxt:109-112 filler
+                    nn.init.constant_(m.bias, bias_v)  # def.prototxt:109-112 filler
             elif isinstance(m, (ConvBiasReLU, Conv1x1BiasReLU)):
-                nn.init.constant_(m.bias, 0.2)  # same filler, bias lives outside conv
+                nn.init.constant_(m.bias, bias_v)  # same filler, bias outside conv
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = self.norm1(self.pool1(self.conv1(x)))

@@ -38,6 +38,11 @@ def main(argv=None):
                    help="per-phase HIP-event timing (eager steps only)")
     p.add_argument("--base-lr", type=float, default=None,
                    help="override solver base_lr (e.g. warmup demos)")
+    p.add_argument("--init", choices=["caffe", "modern"], default="caffe",
+                   help="weight init: caffe = the reference's exact filler "
+                        "(xavier + 0.2 bias; starts collapsed — the reference "
+                        "schedules 2M iters), modern = kaiming/zero-bias "
+                        "(documented deviation; converges from iter 1)")
     p.add_argument("--display", type=int, default=None,
                    help="override solver display interval")
     args = p.parse_args(argv)
@@ -73,7 +78,8 @@ def main(argv=None):
         net_text, solver, synthetic_classes=args.synthetic_classes,
         amp_dtype=amp_dtype, caffemodel=caffemodel,
         num_workers=args.num_workers, pure_bf16=args.pure_bf16,
-        backbone=args.backbone, hip_graph=use_graph, timers=args.timers)
+        backbone=args.backbone, hip_graph=use_graph, timers=args.timers,
+        init=args.init)
     if args.weights and args.weights.endswith(".pt"):
         trainer.restore(args.weights)
     trainer.fit(max_iter=args.max_iter)
