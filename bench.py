@@ -303,6 +303,9 @@ def main():
                 "parallelism": f"dp{world}",
                 "hip_graph": graph is not None,
                 "data_pipeline": pipe is not None,
+                # 288 GB HBM3E sizing evidence (BASELINE config 5)
+                "peak_mem_gb": round(torch.cuda.max_memory_allocated() / 2**30, 2)
+                               if use_cuda else None,
                 "recall_top1_last_step": float(last_out["top1"]) if last_out else None,
             },
         }
