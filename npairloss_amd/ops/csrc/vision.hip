@@ -490,6 +490,87 @@ __global__ void maxpool3_bwd_kernel(const T* __restrict__ dy,
 }
 
 // ---------------------------------------------------------------------------
+// stride-1 rolling-column FORWARD (NHWC, V=4): one thread walks a whole
+// output row (b, oh, cv).  Consecutive 3x3 windows overlap in 2 columns:
+// a 3-column register ring (3 rows x 4 channels) shifts left each step and
+// loads only the NEW column — 1/3 the taps of the per-element kernel.
+// ---------------------------------------------------------------------------
+
+struct MpXCol {
+  float v[3][4];  // x values for the 3 window rows x 4 channels
+};
+
+template <typename T>
+DEVINL void mp_load_xcol(MpXCol& c, const T* __restrict__ x, int w, int W,
+                         long long xsw, const long long xrow[3],
+                         const int rvalid[3]) {
+#pragma unroll
+  for (int rr = 0; rr < 3; ++rr) {
+    if (rvalid[rr] && w >= 0 && w < W) {
+      ld4(x + xrow[rr] + (long long)w * xsw, c.v[rr]);
+    } else {
+      c.v[rr][0] = c.v[rr][1] = c.v[rr][2] = c.v[rr][3] = -FLT_MAX;
+    }
+  }
+}
+
+template <typename T>
+__global__ void maxpool3_fwd_s1_row_kernel(const T* __restrict__ x,
+                                           T* __restrict__ y,
+                                           unsigned char* __restrict__ idx,
+                                           int B, int Cv, int H, int W,
+                                           int OH, int OW,
+                                           long long xsb, long long xsh,
+                                           long long ysb, long long ysh) {
+  const unsigned int total = (unsigned int)((long long)B * OH * Cv);
+  for (unsigned int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    const int cv = i % (unsigned int)Cv;
+    const unsigned int r = i / (unsigned int)Cv;
+    const int oh = r % (unsigned int)OH;
+    const int b = r / (unsigned int)OH;
+    long long xrow[3];
+    int rvalid[3];
+#pragma unroll
+    for (int rr = 0; rr < 3; ++rr) {
+      const int h = oh - 1 + rr;
+      rvalid[rr] = (h >= 0 && h < H);
+      xrow[rr] = (long long)b * xsb + (long long)h * xsh + (long long)(cv * 4);
+    }
+    const long long xsw = (long long)Cv * 4;  // NHWC w-stride = C
+    MpXCol colA, colB, colC;  // w = ow-1, ow, ow+1
+    mp_load_xcol(colA, x, -1, W, xsw, xrow, rvalid);
+    mp_load_xcol(colB, x, 0, W, xsw, xrow, rvalid);
+    long long yi = (long long)b * ysb + (long long)oh * ysh + (long long)(cv * 4);
+    for (int ow = 0; ow < OW; ++ow, yi += (long long)Cv * 4) {
+      mp_load_xcol(colC, x, ow + 1, W, xsw, xrow, rvalid);
+      float best[4];
+      int besti[4];
+#pragma unroll
+      for (int v = 0; v < 4; ++v) {
+        best[v] = -FLT_MAX;
+        besti[v] = 0;
+      }
+#pragma unroll
+      for (int rr = 0; rr < 3; ++rr) {
+#pragma unroll
+        for (int v = 0; v < 4; ++v) {
+          if (colA.v[rr][v] > best[v]) { best[v] = colA.v[rr][v]; besti[v] = rr * 3 + 0; }
+          if (colB.v[rr][v] > best[v]) { best[v] = colB.v[rr][v]; besti[v] = rr * 3 + 1; }
+          if (colC.v[rr][v] > best[v]) { best[v] = colC.v[rr][v]; besti[v] = rr * 3 + 2; }
+        }
+      }
+      st4(y + yi, best);
+      uchar4 u;
+      u.x = besti[0]; u.y = besti[1]; u.z = besti[2]; u.w = besti[3];
+      *reinterpret_cast<uchar4*>(idx + yi) = u;
+      colA = colB;
+      colB = colC;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // stride-1 rolling-column backward (NHWC, V=4): one thread walks a whole
 // input row (b, h, cv).  The <=9 covering windows of consecutive w overlap
 // in 6 columns, so instead of re-gathering 9 (idx, dy) pairs per element
@@ -571,6 +652,78 @@ __global__ void maxpool3_bwd_s1_row_kernel(const T* __restrict__ dy,
       st4(dx + xi, acc);
       colA = colB;
       colB = colC;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// stride-2 rolling-column backward (NHWC, V=4): one thread walks an input
+// row.  An input (h, w) is covered by <=2x2 windows: oh in
+// [ceil((h-1)/2), floor((h+1)/2)] (fixed per thread), ow in
+// [ceil((w-1)/2), floor((w+1)/2)] — the ow window advances every OTHER w
+// step, so a 2-column ring loads one new column per TWO inputs (1/4 the
+// gather traffic of the per-element kernel).
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void maxpool3_bwd_s2_row_kernel(const T* __restrict__ dy,
+                                           const unsigned char* __restrict__ idx,
+                                           T* __restrict__ dx,
+                                           int B, int Cv, int H, int W,
+                                           int OH, int OW,
+                                           long long xsb, long long xsh,
+                                           long long ysb, long long ysh) {
+  const unsigned int total = (unsigned int)((long long)B * H * Cv);
+  for (unsigned int i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * blockDim.x) {
+    const int cv = i % (unsigned int)Cv;
+    const unsigned int r = i / (unsigned int)Cv;
+    const int h = r % (unsigned int)H;
+    const int b = r / (unsigned int)H;
+    // covering output rows: oh0 = ceil((h-1)/2), oh1 = floor((h+1)/2)
+    const int oh0 = (h == 0) ? 0 : (h - 1 + 1) / 2;  // ceil((h-1)/2), h>=0
+    const int oh1 = (h + 1) / 2;
+    long long yrow[2];
+    int rvalid[2];
+#pragma unroll
+    for (int rr = 0; rr < 2; ++rr) {
+      const int oh = oh0 + rr;
+      rvalid[rr] = (oh <= oh1 && oh >= 0 && oh < OH);
+      yrow[rr] = (long long)b * ysb + (long long)oh * ysh + (long long)(cv * 4);
+    }
+    const long long ysw = (long long)Cv * 4;
+    // ring: colP = ow k-1, colQ = ow k with k = floor((w+1)/2)
+    MpCol<T> colP, colQ;
+    mp_load_col(colP, dy, idx, -1, OW, ysw, yrow, rvalid);  // k-1 at w=0 is -1... k(0)=0
+    mp_load_col(colQ, dy, idx, 0, OW, ysw, yrow, rvalid);
+    int k = 0;
+    long long xi = (long long)b * xsb + (long long)h * xsh + (long long)(cv * 4);
+    for (int w = 0; w < W; ++w, xi += (long long)Cv * 4) {
+      const int knew = (w + 1) >> 1;  // floor((w+1)/2)
+      if (knew != k) {
+        colP = colQ;
+        mp_load_col(colQ, dy, idx, knew, OW, ysw, yrow, rvalid);
+        k = knew;
+      }
+      // cols covering w: even w -> only ow = k = w/2 with dw = 1;
+      // odd w -> ow = k-1 (colP, dw = 2) and ow = k (colQ, dw = 0).
+      // Invalid rows carry id bytes 0xFF which match no want code.
+      const bool odd = (w & 1) != 0;
+      const int dwQ = odd ? 0 : 1;
+      float acc[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int rr = 0; rr < 2; ++rr) {
+        const int dh = h - ((oh0 + rr) * 2 - 1);
+        const unsigned int wantQ = (unsigned int)(dh * 3 + dwQ);
+        const unsigned int wantP = (unsigned int)(dh * 3 + 2);
+#pragma unroll
+        for (int v = 0; v < 4; ++v) {
+          const int sh = 8 * v;
+          if (((colQ.id[rr] >> sh) & 0xFFu) == wantQ) acc[v] += colQ.g[rr][v];
+          if (odd && ((colP.id[rr] >> sh) & 0xFFu) == wantP) acc[v] += colP.g[rr][v];
+        }
+      }
+      st4(dx + xi, acc);
     }
   }
 }
@@ -780,7 +933,15 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
               "maxpool3: tensor exceeds the 32-bit index fast path");
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(x, "maxpool3_fwd", {
-    if (V == 4)
+    if (V == 4 && stride == 1)
+      // rolling-column row walker: 1/3 the tap loads (thread owns an
+      // output row; consecutive windows share 2 of 3 columns)
+      maxpool3_fwd_s1_row_kernel<T><<<grid_for((long long)B * OH * (C / 4)),
+                                      NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
+          idx.data_ptr<unsigned char>(), B, C / 4, H, W, OH, OW,
+          xsb, xsh, ysb, ysh);
+    else if (V == 4)
       maxpool3_fwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
           idx.data_ptr<unsigned char>(), B, C / 4, H, W, OH, OW, (int)stride,
@@ -818,6 +979,14 @@ torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
       // rolling-column row walker: 1/3 the gather traffic of the generic
       // per-element kernel (each thread owns a (b, h, cv) input row)
       maxpool3_bwd_s1_row_kernel<T><<<grid_for((long long)B * H * (C / 4)),
+                                      NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
+          reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW,
+          xsb, xsh, ysb, ysh);
+    else if (V == 4 && stride == 2)
+      // 2-column ring: one new (idx, dy) column per TWO inputs — 1/4 the
+      // gather traffic of the generic kernel
+      maxpool3_bwd_s2_row_kernel<T><<<grid_for((long long)B * H * (C / 4)),
                                       NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
           reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW,

@@ -170,9 +170,10 @@ def test_recall_vs_torch(bg):
     ks = [1, 5, 10]
     ks_t = torch.tensor(ks, dtype=torch.int32, device="cuda")
     hits = _C().recall_hits(S, lab_l.int(), lab_g.int(), 0, ks_t, max(ks))
-    ref = NL._recall_torch(S, lab_l, lab_g, 0, ks) * B
-    # exact: same threshold extraction + strict-> semantics (tie fixtures
-    # with deliberate duplicates live in test_gpu_ties.py)
+    # exact integer hit counts: same threshold extraction + strict->
+    # semantics (tie fixtures with deliberate duplicates: test_gpu_ties.py).
+    # round() undoes the reference's fp32 /B*B wobble (e.g. 77/120*120)
+    ref = torch.round(NL._recall_torch(S, lab_l, lab_g, 0, ks) * B)
     torch.testing.assert_close(hits.float(), ref, atol=0, rtol=0)
 
 

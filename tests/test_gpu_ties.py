@@ -40,7 +40,7 @@ def test_recall_with_ties_exact(bg):
     ks = [1, 5, 10]
     ks_t = torch.tensor(ks, dtype=torch.int32, device="cuda")
     hits = _C().recall_hits(S, lab_l.int(), lab_g.int(), 0, ks_t, max(ks))
-    ref = NL._recall_torch(S, lab_l, lab_g, 0, ks) * B
+    ref = torch.round(NL._recall_torch(S, lab_l, lab_g, 0, ks) * B)
     torch.testing.assert_close(hits.float(), ref, atol=0, rtol=0)
 
 
