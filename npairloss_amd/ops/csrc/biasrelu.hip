@@ -209,7 +209,7 @@ __global__ void biasrelu_db_finalize_kernel(const float* __restrict__ partials,
 static void br_db_reduce(const torch::Tensor& partials, torch::Tensor& db,
                          int C, int nblocks, hipStream_t stream) {
   constexpr int NCHUNK = 16;
-  if (nblocks <= 512) {
+  if (nblocks <= 64) {
     biasrelu_db_finalize_kernel<<<(C + 255) / 256, 256, 0, stream>>>(
         partials.data_ptr<float>(), db.data_ptr<float>(), C, nblocks);
     return;
@@ -296,10 +296,8 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
     if (nhwc && C % V == 0 && C / V <= NPAIR_BLOCK) {
       const long long groups = total / V;
       // smaller grid than the forward: the cross-block finalize reads
-      // grid*C partials serially per channel, so fewer/larger blocks win —
-      // at <=512 a SINGLE one-launch finalize suffices (the 2-stage chunk
-      // path cost ~1 ms/step in extra-launch latency at 57 convs/step)
-      const int grid = br_fixed_grid(groups, (int)(C / V), 512);
+      // grid*C partials serially per channel, so fewer/larger blocks win
+      const int grid = br_fixed_grid(groups, (int)(C / V), 1280);
       auto partials = torch::empty({grid, C}, y.options().dtype(torch::kFloat32));
       biasrelu_bwd_vec_kernel<T><<<grid, NPAIR_BLOCK,
                                    (size_t)NPAIR_BLOCK * V * sizeof(float), stream>>>(

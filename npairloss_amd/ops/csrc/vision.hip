@@ -933,15 +933,10 @@ std::vector<torch::Tensor> maxpool3_fwd(torch::Tensor x, int64_t stride, bool ce
               "maxpool3: tensor exceeds the 32-bit index fast path");
   auto stream = at::hip::getCurrentHIPStream();
   VISION_DISPATCH(x, "maxpool3_fwd", {
-    if (V == 4 && stride == 1)
-      // rolling-column row walker: 1/3 the tap loads (thread owns an
-      // output row; consecutive windows share 2 of 3 columns)
-      maxpool3_fwd_s1_row_kernel<T><<<grid_for((long long)B * OH * (C / 4)),
-                                      NPAIR_BLOCK, 0, stream>>>(
-          reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
-          idx.data_ptr<unsigned char>(), B, C / 4, H, W, OH, OW,
-          xsb, xsh, ysb, ysh);
-    else if (V == 4)
+    // note: a rolling-column s1 forward (maxpool3_fwd_s1_row_kernel) was
+    // measured SLOWER than the per-element kernel — the serial row walk
+    // loses too much parallelism at the 14^2/7^2 inception shapes
+    if (V == 4)
       maxpool3_fwd_kernel<T, 4><<<grid_for(total), NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(xc.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
           idx.data_ptr<unsigned char>(), B, C / 4, H, W, OH, OW, (int)stride,
@@ -979,14 +974,6 @@ torch::Tensor maxpool3_bwd(torch::Tensor dy, torch::Tensor idx, int64_t stride,
       // rolling-column row walker: 1/3 the gather traffic of the generic
       // per-element kernel (each thread owns a (b, h, cv) input row)
       maxpool3_bwd_s1_row_kernel<T><<<grid_for((long long)B * H * (C / 4)),
-                                      NPAIR_BLOCK, 0, stream>>>(
-          reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
-          reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW,
-          xsb, xsh, ysb, ysh);
-    else if (V == 4 && stride == 2)
-      // 2-column ring: one new (idx, dy) column per TWO inputs — 1/4 the
-      // gather traffic of the generic kernel
-      maxpool3_bwd_s2_row_kernel<T><<<grid_for((long long)B * H * (C / 4)),
                                       NPAIR_BLOCK, 0, stream>>>(
           reinterpret_cast<const T*>(dyc.data_ptr()), idx.data_ptr<unsigned char>(),
           reinterpret_cast<T*>(dx.data_ptr()), B, C / 4, (int)H, (int)W, OH, OW,
