@@ -79,12 +79,18 @@ def main():
         torch.backends.cudnn.benchmark = True  # MIOpen find-best for fixed shapes
 
     if world > 1:
+        import datetime
+
         import torch.distributed as dist
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29517")
+        # failure detection: a hung peer fails the collective after this
+        # timeout instead of deadlocking the job (reference: MPI = job hang)
+        tmo = int(os.environ.get("NPAIR_COMM_TIMEOUT_S", "300"))
         dist.init_process_group("nccl" if use_cuda else "gloo",
-                                rank=rank, world_size=world)
+                                rank=rank, world_size=world,
+                                timeout=datetime.timedelta(seconds=tmo))
 
     from npairloss_amd.config.params import NPairLossConfig
     from npairloss_amd.models import build_embedding_model

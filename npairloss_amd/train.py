@@ -63,10 +63,14 @@ def main(argv=None):
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     if world > 1:
+        import datetime
+
         import torch.distributed as dist
 
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo")
+        tmo = int(os.environ.get("NPAIR_COMM_TIMEOUT_S", "300"))
+        dist.init_process_group("nccl" if torch.cuda.is_available() else "gloo",
+                                timeout=datetime.timedelta(seconds=tmo))
         if torch.cuda.is_available():
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
 
