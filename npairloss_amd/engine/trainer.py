@@ -109,6 +109,9 @@ class Trainer:
         self.model.load_state_dict(ck["model"])
         self.optimizer.load_state_dict(ck["optimizer"])
         self.iter = ck["iter"]
+        # optimizer load_state_dict REPLACES state tensors; a captured graph
+        # would keep updating the orphaned ones — force recapture
+        self._graph_state = None
 
     # -- one training iteration --------------------------------------------
 
@@ -219,6 +222,8 @@ class Trainer:
                                          lab=static_lab, out=out, lr=lr)
             self.iter += 1
             return self._stats_from(g, lr)
+        if images.shape != g["img"].shape or labels.shape != g["lab"].shape:
+            return self.train_step(images, labels)  # odd-sized batch: eager
         g["img"].copy_(self._prepare(images.to(self.device, non_blocking=True)))
         g["lab"].copy_(labels.to(self.device, non_blocking=True))
         g["graph"].replay()

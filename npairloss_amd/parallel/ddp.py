@@ -58,9 +58,11 @@ class BucketedGradReducer:
         self._views: Dict[int, torch.Tensor] = {}
         self._accumulate = False
         # flat gradient buffers pay off even at world=1 (fused zero, fixed
-        # addresses for hipGraph); comm only happens when enabled.
-        self.use_flat = flat_grads
-        if not self.use_flat and not self.enabled:
+        # addresses for hipGraph); comm only happens when enabled.  With
+        # comm enabled the flats are MANDATORY (the all-reduce runs on
+        # them, so grads must live there).
+        self.use_flat = flat_grads or self.enabled
+        if not self.use_flat:
             return
         # reverse order: later layers' grads arrive first during backward.
         # A bucket never mixes dtypes (one flat buffer each).
