@@ -24,8 +24,6 @@
 #include <hip/hip_bf16.h>
 
 #include <numeric>
-#include <mutex>
-#include <unordered_map>
 
 #include "common.h"
 
@@ -87,18 +85,12 @@ __global__ void biasrelu_fwd_vec_kernel(const T* __restrict__ x,
 
 // backward: dx = dy * (y > 0); db[c] = ordered sum of dx over sites.
 // Each thread owns channels [c0, c0+V); register accumulate, then an
-// ordered per-block LDS reduce to partials, then the LAST block (counter
-// handshake) sums partials over blocks in ascending order straight into
-// db — ONE launch, still bitwise-deterministic (the atomic only elects
-// the last block; every float add has a fixed order).  The counter
-// self-resets so the buffer is reusable without a zeroing launch.
+// ordered per-block LDS reduce, then the ordered cross-block finalize.
 template <typename T>
 __global__ void biasrelu_bwd_vec_kernel(const T* __restrict__ y,
                                         const T* __restrict__ dy,
                                         T* __restrict__ dx,
                                         float* __restrict__ partials,
-                                        unsigned int* __restrict__ counter,
-                                        float* __restrict__ db,
                                         long long groups, int C) {
   constexpr int V = Pack16<T>::N;
   extern __shared__ float lds[];  // blockDim * V floats
@@ -146,21 +138,6 @@ __global__ void biasrelu_bwd_vec_kernel(const T* __restrict__ y,
 #pragma unroll
     for (int k = 0; k < V; ++k) out[k] = s[k];
   }
-  // last-block cross-block finalize (ascending block order — deterministic)
-  __shared__ unsigned int am_last;
-  __threadfence();
-  __syncthreads();
-  if (threadIdx.x == 0)
-    am_last = (atomicAdd(counter, 1u) == (unsigned int)gridDim.x - 1u) ? 1u : 0u;
-  __syncthreads();
-  if (!am_last) return;
-  __threadfence();  // make every block's partials visible
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    float s = 0.f;
-    for (int b = 0; b < (int)gridDim.x; ++b) s += partials[(long long)b * C + c];
-    db[c] = s;
-  }
-  if (threadIdx.x == 0) *counter = 0u;  // rearm for the next call
 }
 
 // ---------------------------------------------------------------------------
@@ -255,22 +232,6 @@ static bool br_nhwc(const torch::Tensor& t) {
   return t.is_contiguous(at::MemoryFormat::ChannelsLast);
 }
 
-// per-device zero-initialized counter for the last-block handshake; the
-// kernel rearms it to 0, and all uses share one compute stream (launches
-// serialize), so one cached tensor per device suffices.  Fixed address —
-// hipGraph-capture safe.
-static torch::Tensor br_counter(const torch::Device& dev) {
-  static std::unordered_map<int, torch::Tensor> cache;
-  static std::mutex mu;
-  std::lock_guard<std::mutex> g(mu);
-  auto it = cache.find((int)dev.index());
-  if (it == cache.end()) {
-    auto t = torch::zeros({1}, torch::TensorOptions().dtype(torch::kInt32).device(dev));
-    it = cache.emplace((int)dev.index(), t).first;
-  }
-  return it->second;
-}
-
 // grid size such that (grid * NPAIR_BLOCK) % q == 0 (q = C/VEC): pins every
 // thread's channel window across grid-stride iterations
 static int br_fixed_grid(long long groups, int q, int maxgrid) {
@@ -338,14 +299,13 @@ std::vector<torch::Tensor> biasrelu_bwd(torch::Tensor y, torch::Tensor dy) {
       // grid*C partials serially per channel, so fewer/larger blocks win
       const int grid = br_fixed_grid(groups, (int)(C / V), 1280);
       auto partials = torch::empty({grid, C}, y.options().dtype(torch::kFloat32));
-      auto counter = br_counter(y.device());
       biasrelu_bwd_vec_kernel<T><<<grid, NPAIR_BLOCK,
                                    (size_t)NPAIR_BLOCK * V * sizeof(float), stream>>>(
           reinterpret_cast<const T*>(yc.data_ptr()),
           reinterpret_cast<const T*>(dyc.data_ptr()),
           reinterpret_cast<T*>(dx.data_ptr()), partials.data_ptr<float>(),
-          reinterpret_cast<unsigned int*>(counter.data_ptr<int>()),
-          db.data_ptr<float>(), groups, (int)C);
+          groups, (int)C);
+      br_db_reduce(partials, db, (int)C, grid, stream);
     } else {
       const int blocks = (int)std::min<long long>((total + NPAIR_BLOCK - 1) / NPAIR_BLOCK, 1024);
       auto partials = torch::empty({blocks, C}, y.options().dtype(torch::kFloat32));
