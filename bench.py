@@ -250,7 +250,18 @@ def main():
             graph = None
             print(f"hipGraph capture failed ({e!r}); running eager", file=sys.stderr)
         if world > 1:
+            # consensus: replaying on some ranks while others run eager
+            # keeps the collective SEQUENCE identical, but a rank whose
+            # capture failed mid-way may have desynced RCCL state — if any
+            # rank failed, every rank falls back to eager
             import torch.distributed as dist
+            ok = torch.tensor([0 if graph is None else 1],
+                              device=device if use_cuda else "cpu")
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if int(ok.item()) == 0 and graph is not None:
+                graph = None
+                print("peer rank failed hipGraph capture; running eager",
+                      file=sys.stderr)
             dist.barrier()
 
     def run_step(i):

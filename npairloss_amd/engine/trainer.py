@@ -18,6 +18,7 @@ import os
 import time
 from typing import Callable, Iterable, Optional
 
+import numpy as np
 import torch
 
 from ..config.params import SolverConfig
@@ -112,6 +113,72 @@ class Trainer:
         # optimizer load_state_dict REPLACES state tensors; a captured graph
         # would keep updating the orphaned ones — force recapture
         self._graph_state = None
+
+    # -- Caffe-format snapshots (.caffemodel + .solverstate) ----------------
+
+    def snapshot_caffe(self, prefix: Optional[str] = None) -> Optional[tuple]:
+        """Write `<prefix>iter_N.caffemodel` (weights by layer name) plus
+        `<prefix>iter_N.solverstate` (iter + SGD momentum history in
+        canonical learnable-parameter order), like the reference solver
+        (usage/solver.prototxt:15-16)."""
+        if comm.rank() != 0:
+            return None
+        from ..utils.caffemodel import (CaffeSolverState, caffe_param_order,
+                                        save_caffemodel, write_solverstate)
+
+        prefix = prefix or self.solver.snapshot_prefix or "./snap/model_"
+        os.makedirs(os.path.dirname(prefix) or ".", exist_ok=True)
+        mpath = f"{prefix}iter_{self.iter}.caffemodel"
+        spath = f"{prefix}iter_{self.iter}.solverstate"
+        save_caffemodel(self.model, mpath)
+        history = []
+        for _, p in caffe_param_order(self.model):
+            v = self.optimizer.state.get(p, {}).get("v")
+            if v is None:
+                v = torch.zeros_like(p, dtype=torch.float32)
+            # .contiguous(): channels_last tensors serialize in logical
+            # (NCHW) element order, matching Caffe blob storage
+            history.append(v.detach().float().cpu().contiguous().numpy())
+        write_solverstate(spath, CaffeSolverState(
+            iter=self.iter, learned_net=os.path.basename(mpath),
+            history=history))
+        return mpath, spath
+
+    def restore_caffe(self, solverstate_path: str,
+                      caffemodel_path: Optional[str] = None):
+        """Resume mid-training from Caffe artifacts: weights from the
+        .caffemodel (name-matched), iteration + SGD momentum from the
+        .solverstate (canonical parameter order)."""
+        from ..utils.caffemodel import (caffe_param_order, load_caffemodel_into,
+                                        read_solverstate)
+
+        st = read_solverstate(solverstate_path)
+        if caffemodel_path is None and st.learned_net:
+            cand = os.path.join(os.path.dirname(solverstate_path), st.learned_net)
+            caffemodel_path = cand if os.path.exists(cand) else None
+        if caffemodel_path:
+            load_caffemodel_into(self.model, caffemodel_path)
+        params = caffe_param_order(self.model)
+        if len(st.history) != len(params):
+            raise ValueError(
+                f"solverstate history has {len(st.history)} blobs, model has "
+                f"{len(params)} learnable parameters")
+        with torch.no_grad():
+            for (name, p), h in zip(params, st.history):
+                if h.size != p.numel():
+                    raise ValueError(f"history blob for {name}: {h.size} elements "
+                                     f"vs parameter {p.numel()}")
+                stp = self.optimizer.state.setdefault(p, {})
+                master = stp.get("master")
+                ref = master if master is not None else p
+                t = torch.from_numpy(np.ascontiguousarray(h)).reshape(
+                    p.shape).to(device=ref.device, dtype=ref.dtype)
+                if (t.dim() == 4 and not ref.is_contiguous()
+                        and ref.is_contiguous(memory_format=torch.channels_last)):
+                    t = t.contiguous(memory_format=torch.channels_last)
+                stp["v"] = t
+        self.iter = st.iter
+        self._graph_state = None  # state tensors replaced: recapture
 
     # -- one training iteration --------------------------------------------
 
@@ -216,6 +283,16 @@ class Trainer:
                     out = self._compute_step(static_img, static_lab)
             except Exception as e:  # noqa: BLE001
                 self.log(f"hipGraph capture failed ({e!r}); continuing eager")
+                graph = None
+            if comm.is_dist() and comm.world_size() > 1:
+                # consensus: if any rank failed capture, all ranks go eager
+                # (a mid-capture failure may have desynced RCCL state)
+                import torch.distributed as dist
+                ok = torch.tensor([0 if graph is None else 1], device=self.device)
+                dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+                if int(ok.item()) == 0:
+                    graph = None
+            if graph is None:
                 self.use_graph = False
                 return self.train_step(images, labels)
             self._graph_state = g = dict(graph=graph, img=static_img,

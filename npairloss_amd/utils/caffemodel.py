@@ -216,6 +216,76 @@ def write_caffemodel(path: str, layers: List[CaffeLayer], net_name: str = "net")
 
 
 # ---------------------------------------------------------------------------
+# SolverState (.solverstate): mid-training resume artifacts
+#
+#   SolverState { iter=1 (int32); learned_net=2 (string);
+#                 history=3 (BlobProto, repeated); current_step=4 (int32) }
+#
+# `history` holds one blob per learnable parameter — the previous SGD
+# update vector (Caffe's momentum state; identical quantity to CaffeSGD's
+# per-param "v") — in net learnable-parameter order: for each layer of
+# caffe_names(), weight then bias.
+# ---------------------------------------------------------------------------
+
+
+@dataclass
+class CaffeSolverState:
+    iter: int = 0
+    learned_net: str = ""
+    history: List[np.ndarray] = field(default_factory=list)
+    current_step: int = 0
+
+
+def read_solverstate(path_or_bytes) -> CaffeSolverState:
+    if isinstance(path_or_bytes, (bytes, bytearray)):
+        buf = bytes(path_or_bytes)
+    else:
+        with open(path_or_bytes, "rb") as fh:
+            buf = fh.read()
+    st = CaffeSolverState()
+    for fno, wt, v in _iter_fields(buf):
+        if fno == 1 and wt == 0:
+            st.iter = int(v)
+        elif fno == 2 and wt == 2:
+            st.learned_net = v.decode("utf-8", "replace")
+        elif fno == 3 and wt == 2:
+            st.history.append(_parse_blob(v))
+        elif fno == 4 and wt == 0:
+            st.current_step = int(v)
+    return st
+
+
+def write_solverstate(path: str, state: CaffeSolverState) -> None:
+    out = bytearray()
+    out += _key(1, 0)
+    _write_varint(out, int(state.iter))
+    if state.learned_net:
+        out += _len_delim(2, state.learned_net.encode())
+    for h in state.history:
+        out += _len_delim(3, _encode_blob(np.asarray(h, dtype=np.float32)))
+    out += _key(4, 0)
+    _write_varint(out, int(state.current_step))
+    with open(path, "wb") as fh:
+        fh.write(bytes(out))
+
+
+def caffe_param_order(model: torch.nn.Module) -> List[Tuple[str, torch.Tensor]]:
+    """Canonical learnable-parameter order for solverstate history blobs:
+    caffe_names() layer order, weight then bias per layer."""
+    target = model
+    if not hasattr(target, "caffe_names") and hasattr(target, "backbone"):
+        target = target.backbone
+    if not hasattr(target, "caffe_names"):
+        raise TypeError("model does not expose caffe_names()")
+    out: List[Tuple[str, torch.Tensor]] = []
+    for name, mod in target.caffe_names().items():
+        out.append((f"{name}/weight", mod.weight))
+        if getattr(mod, "bias", None) is not None:
+            out.append((f"{name}/bias", mod.bias))
+    return out
+
+
+# ---------------------------------------------------------------------------
 # model loading
 # ---------------------------------------------------------------------------
 
