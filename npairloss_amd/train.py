@@ -86,6 +86,11 @@ def main(argv=None):
         init=args.init)
     if args.weights and args.weights.endswith(".pt"):
         trainer.restore(args.weights)
+    elif args.weights and args.weights.endswith(".solverstate"):
+        # mid-training resume from Caffe artifacts (weights via the
+        # learned_net reference next to the .solverstate, momentum + iter
+        # from the state blobs)
+        trainer.restore_caffe(args.weights)
     trainer.fit(max_iter=args.max_iter)
 
 
