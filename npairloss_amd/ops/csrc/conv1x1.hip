@@ -20,6 +20,8 @@
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
+
+#include <cstdlib>
 #include <hip/hip_bf16.h>
 
 #include "common.h"
@@ -116,6 +118,146 @@ __global__ void conv1x1_nt_kernel(const unsigned short* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------------
+// v2: 128x64 tile, BK=64, register-prefetch double buffering, +8-half LDS
+// row padding (the 64-half pitch put all 16 A-frag lanes on one bank
+// quartet).  4 waves in a 2x2 grid, each computing 64x32 via 4x2
+// fragments -> 16 MFMAs per K-tile between barriers (v1: 4).
+// ---------------------------------------------------------------------------
+
+#define C2_BM 128
+#define C2_BN 64
+#define C2_BK 64
+#define C2_PAD 8  // halves
+
+template <int EPI>
+__launch_bounds__(NPAIR_BLOCK)
+__global__ void conv1x1_nt_v2_kernel(const unsigned short* __restrict__ A,
+                                     const unsigned short* __restrict__ B,
+                                     const float* __restrict__ bias,
+                                     unsigned short* __restrict__ Y,
+                                     int M, int N, int K) {
+  __shared__ unsigned short As[C2_BM][C2_BK + C2_PAD];
+  __shared__ unsigned short Bs[C2_BN][C2_BK + C2_PAD];
+  const int m0 = blockIdx.y * C2_BM;
+  const int n0 = blockIdx.x * C2_BN;
+  const int t = threadIdx.x;
+  const int wid = t / WAVE;
+  const int lane = t % WAVE;
+  const int wm = (wid >> 1) * 64;  // 2x2 waves: wave tile 64 x 32
+  const int wn = (wid & 1) * 32;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+
+  // stage coordinates: 256 threads x 32 halves = one 128x64 A tile pass
+  const int ar = t >> 1;             // A row 0..127
+  const int ac = (t & 1) * 32;       // A col 0 or 32
+  const int br = t >> 2;             // B row 0..63
+  const int bc = (t & 3) * 16;       // B col 0,16,32,48
+
+  f32x4 acc[4][2] = {};
+
+  // prefetch registers for the NEXT K-tile (4+2 16B vectors)
+  using v8 = bf16x8;
+  v8 pa[4], pb[2];
+
+  auto load_a = [&](int k0, v8* dst) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = ar;
+      const int col = ac + i * 8;
+      const int gm = m0 + row;
+      const int gk = k0 + col;
+      if (gm < M && gk + 7 < K) {
+        dst[i] = *reinterpret_cast<const v8*>(&A[(size_t)gm * K + gk]);
+      } else {
+        unsigned short tmp[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          tmp[e] = (gm < M && gk + e < K) ? A[(size_t)gm * K + gk + e] : 0;
+        dst[i] = *reinterpret_cast<v8*>(tmp);
+      }
+    }
+  };
+  auto load_b = [&](int k0, v8* dst) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int row = br;
+      const int col = bc + i * 8;
+      const int gn = n0 + row;
+      const int gk = k0 + col;
+      if (gn < N && gk + 7 < K) {
+        dst[i] = *reinterpret_cast<const v8*>(&B[(size_t)gn * K + gk]);
+      } else {
+        unsigned short tmp[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          tmp[e] = (gn < N && gk + e < K) ? B[(size_t)gn * K + gk + e] : 0;
+        dst[i] = *reinterpret_cast<v8*>(tmp);
+      }
+    }
+  };
+  auto store_stage = [&]() {
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      *reinterpret_cast<v8*>(&As[ar][ac + i * 8]) = pa[i];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      *reinterpret_cast<v8*>(&Bs[br][bc + i * 8]) = pb[i];
+  };
+
+  load_a(0, pa);
+  load_b(0, pb);
+  store_stage();
+  __syncthreads();
+
+  for (int k0 = 0; k0 < K; k0 += C2_BK) {
+    const bool last = (k0 + C2_BK >= K);
+    if (!last) {  // prefetch next tile into registers during compute
+      load_a(k0 + C2_BK, pa);
+      load_b(k0 + C2_BK, pb);
+    }
+#pragma unroll
+    for (int kk = 0; kk < C2_BK; kk += 32) {
+      const int kf = kk + l4 * 8;
+#pragma unroll
+      for (int fm = 0; fm < 4; ++fm) {
+        const v8 a = *reinterpret_cast<const v8*>(&As[wm + fm * 16 + l15][kf]);
+#pragma unroll
+        for (int fn = 0; fn < 2; ++fn) {
+          const v8 b = *reinterpret_cast<const v8*>(&Bs[wn + fn * 16 + l15][kf]);
+          acc[fm][fn] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[fm][fn], 0, 0, 0);
+        }
+      }
+    }
+    if (!last) {
+      __syncthreads();
+      store_stage();
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int fm = 0; fm < 4; ++fm)
+#pragma unroll
+    for (int fn = 0; fn < 2; ++fn) {
+      const int gn = n0 + wn + fn * 16 + l15;
+      const float bv = (EPI && gn < N) ? bias[gn] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gm = m0 + wm + fm * 16 + l4 * 4 + r;
+        if (gm < M && gn < N) {
+          float v = acc[fm][fn][r];
+          if (EPI) {
+            v += bv;
+            v = v > 0.f ? v : 0.f;
+          }
+          Y[(size_t)gm * N + gn] = __hip_bfloat16_raw(__float2bfloat16(v)).x;
+        }
+      }
+    }
+}
+
+// ---------------------------------------------------------------------------
 
 static torch::Tensor conv1x1_nt(torch::Tensor A, torch::Tensor B,
                                 c10::optional<torch::Tensor> bias, bool epi) {
@@ -124,7 +266,6 @@ static torch::Tensor conv1x1_nt(torch::Tensor A, torch::Tensor B,
   TORCH_CHECK(A.size(1) == B.size(1), "reduction dims differ");
   const int64_t M = A.size(0), N = B.size(0), K = A.size(1);
   auto Y = torch::empty({M, N}, A.options());
-  dim3 grid((N + C1_BN - 1) / C1_BN, (M + C1_BM - 1) / C1_BM);
   auto stream = at::hip::getCurrentHIPStream();
   const float* bptr = nullptr;
   torch::Tensor bf;
@@ -133,16 +274,32 @@ static torch::Tensor conv1x1_nt(torch::Tensor A, torch::Tensor B,
     bf = bias->to(torch::kFloat32).contiguous();
     bptr = bf.data_ptr<float>();
   }
-  if (epi)
-    conv1x1_nt_kernel<1><<<grid, NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const unsigned short*>(A.data_ptr()),
-        reinterpret_cast<const unsigned short*>(B.data_ptr()), bptr,
-        reinterpret_cast<unsigned short*>(Y.data_ptr()), (int)M, (int)N, (int)K);
-  else
-    conv1x1_nt_kernel<0><<<grid, NPAIR_BLOCK, 0, stream>>>(
-        reinterpret_cast<const unsigned short*>(A.data_ptr()),
-        reinterpret_cast<const unsigned short*>(B.data_ptr()), nullptr,
-        reinterpret_cast<unsigned short*>(Y.data_ptr()), (int)M, (int)N, (int)K);
+  const char* v1 = std::getenv("NPAIR_CONV1X1_V1");
+  if (v1 && v1[0] == '1') {
+    dim3 grid((N + C1_BN - 1) / C1_BN, (M + C1_BM - 1) / C1_BM);
+    if (epi)
+      conv1x1_nt_kernel<1><<<grid, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const unsigned short*>(A.data_ptr()),
+          reinterpret_cast<const unsigned short*>(B.data_ptr()), bptr,
+          reinterpret_cast<unsigned short*>(Y.data_ptr()), (int)M, (int)N, (int)K);
+    else
+      conv1x1_nt_kernel<0><<<grid, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const unsigned short*>(A.data_ptr()),
+          reinterpret_cast<const unsigned short*>(B.data_ptr()), nullptr,
+          reinterpret_cast<unsigned short*>(Y.data_ptr()), (int)M, (int)N, (int)K);
+  } else {
+    dim3 grid((N + C2_BN - 1) / C2_BN, (M + C2_BM - 1) / C2_BM);
+    if (epi)
+      conv1x1_nt_v2_kernel<1><<<grid, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const unsigned short*>(A.data_ptr()),
+          reinterpret_cast<const unsigned short*>(B.data_ptr()), bptr,
+          reinterpret_cast<unsigned short*>(Y.data_ptr()), (int)M, (int)N, (int)K);
+    else
+      conv1x1_nt_v2_kernel<0><<<grid, NPAIR_BLOCK, 0, stream>>>(
+          reinterpret_cast<const unsigned short*>(A.data_ptr()),
+          reinterpret_cast<const unsigned short*>(B.data_ptr()), nullptr,
+          reinterpret_cast<unsigned short*>(Y.data_ptr()), (int)M, (int)N, (int)K);
+  }
   HIP_CHECK_LAST();
   return Y;
 }
