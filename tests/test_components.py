@@ -500,3 +500,24 @@ def test_solverstate_history_length_check(tmp_path):
     assert rt.iter == 7 and rt.learned_net == "x.caffemodel"
     assert rt.current_step == 1
     np.testing.assert_array_equal(rt.history[0], st.history[0])
+
+
+def test_conv1x1_module_modes_cpu():
+    """All three Conv1x1BiasReLU strategies agree on CPU (each falls back
+    to exact eager math there) and backprop to identical gradients."""
+    from npairloss_amd.ops.vision import Conv1x1BiasReLU
+
+    results = []
+    for mode in ("off", "hybrid", "custom"):
+        torch.manual_seed(11)
+        m = Conv1x1BiasReLU(8, 12)
+        m.mode = mode
+        with torch.no_grad():
+            m.bias.uniform_(-0.3, 0.3)
+        x = torch.randn(2, 8, 6, 6, requires_grad=True)
+        y = m(x)
+        y.sum().backward()
+        results.append((y.detach(), x.grad, m.weight.grad, m.bias.grad))
+    for r in results[1:]:
+        for a, b in zip(results[0], r):
+            torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-6)
