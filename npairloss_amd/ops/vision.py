@@ -226,11 +226,13 @@ class _Conv1x1BiasReLUFn(torch.autograd.Function):
 
 
 class _Conv1x1MM(torch.autograd.Function):
-    """Bias-free 1x1 conv with a GEMM backward.  Forward stays on MIOpen
-    (measured faster than both our MFMA kernel and hipBLASLt at the bench
-    shapes — profiles/kb_conv1x1); backward replaces MIOpen's
-    convolution_backward with two hipBLASLt GEMMs on the free NHWC views:
-    dx[M,K] = g[M,N] @ w[N,K],  dw[N,K] = g^T[N,M] @ x[M,K]."""
+    """Bias-free 1x1 conv with a mixed backward.  Forward stays on MIOpen
+    (measured fastest overall at the bench shapes — profiles/kb_conv1x1*);
+    backward: the data gradient runs our v2 MFMA GEMM
+    dx[M,K] = g[M,N] @ (w^T)[K,N]^T (beats hipBLASLt at EVERY bench shape,
+    profiles/kb_conv1x1_v2.log), the weight gradient stays on MIOpen's
+    wgrad-only convolution_backward (an unsplit library TN GEMM loses 10x
+    on the M=800k reduction)."""
 
     @staticmethod
     @torch.amp.custom_fwd(device_type="cuda", cast_inputs=torch.bfloat16)
@@ -251,10 +253,13 @@ class _Conv1x1MM(torch.autograd.Function):
             B, K, H, W = x.shape
             N = w.shape[0]
             gm = g.permute(0, 2, 3, 1).reshape(-1, N)
-            xm = x.permute(0, 2, 3, 1).reshape(-1, K)
             w2 = w.reshape(N, K)
-            dx = (gm @ w2).view(B, H, W, K).permute(0, 3, 1, 2)
-            dw = (gm.t() @ xm).view_as(w)
+            wt = w2.t().contiguous()  # tiny K x N copy
+            dx = (_backend.ext().conv1x1_dgrad(gm, wt)
+                  .view(B, H, W, K).permute(0, 3, 1, 2))
+            dw = torch.ops.aten.convolution_backward(
+                g, x, w, None, (1, 1), (0, 0), (1, 1), False, (0, 0), 1,
+                [False, True, False])[1]
         else:
             dx = F.conv_transpose2d(g, w)
             dw = torch.nn.grad.conv2d_weight(x, w.shape, g)
