@@ -14,7 +14,6 @@ kernels on device, not per-image host code.
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 
 import torch

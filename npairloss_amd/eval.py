@@ -15,7 +15,6 @@ from typing import Dict, Iterable, Sequence, Tuple
 import torch
 
 from .ops import _backend
-from .ops.npair_loss import _recall_torch
 
 
 @torch.no_grad()

@@ -12,7 +12,7 @@ pool -> 1024-d embedding) with the original Caffe layer names preserved in
 
 from __future__ import annotations
 
-from typing import Dict, Tuple
+from typing import Dict
 
 import torch
 from torch import nn
@@ -32,12 +32,14 @@ class ConvReLU(nn.Module):
 
 
 def _conv_block(cin, cout, k, stride=1, pad=0, fused=True):
-    """conv+bias+relu unit.  fused=True (default): 1x1 convs become ONE
-    fused MFMA GEMM + bias + ReLU pass (csrc/conv1x1.hip); other kernels
-    keep bias OUT of the conv and run the fused BiasReLU kernel, whose
-    backward collapses torch's threshold_backward + generic bias-grad
-    reduce into one pass (csrc/biasrelu.hip).  fused=False is the plain
-    conv(bias)+ReLU pair."""
+    """conv+bias+relu unit.  fused=True (default) keeps bias OUT of the
+    conv and runs the fused BiasReLU kernel, whose backward collapses
+    torch's threshold_backward + the generic bias-grad reduce into one
+    pass (csrc/biasrelu.hip).  1x1 convs go through Conv1x1BiasReLU, whose
+    strategy (NPAIR_CONV1X1: off / hybrid / custom) additionally lets the
+    forward or data-gradient run our MFMA GEMM (csrc/conv1x1.hip) — the
+    measured default this round is "off" (MIOpen GEMMs,
+    profiles/kb_conv1x1_v2.log).  fused=False is plain conv(bias)+ReLU."""
     if fused and k == 1 and stride == 1 and pad == 0:
         return Conv1x1BiasReLU(cin, cout)
     if fused:

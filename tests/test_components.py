@@ -1,7 +1,6 @@
 """CPU tests: PK sampler, transforms, caffemodel codec, CaffeSGD, trainer."""
 
 import os
-import tempfile
 
 import numpy as np
 import pytest

@@ -1,6 +1,5 @@
 """Prototxt parsing: the reference's actual usage files must round-trip."""
 
-import math
 import os
 
 from npairloss_amd.config.params import (

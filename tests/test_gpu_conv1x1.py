@@ -4,7 +4,6 @@ pass + dgrad GEMM + hipBLASLt wgrad) against fp32 torch conv references."""
 
 import pytest
 import torch
-import torch.nn.functional as F
 
 pytestmark = [pytest.mark.gpu]
 
@@ -92,8 +91,6 @@ def test_conv1x1_module_end_to_end(amp):
 def test_googlenet_fused_close_to_reference_gpu():
     """Whole GoogLeNet with fused 1x1 GEMM convs vs the unfused build with
     identical weights: embeddings must agree to bf16 tolerance."""
-    import os
-
     from npairloss_amd.models.googlenet import GoogLeNet
 
     torch.manual_seed(3)

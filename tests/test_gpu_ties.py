@@ -5,13 +5,11 @@ picks over lists with repeated values (.cu:282-336), and the mining select
 comparisons at s == thr (.cu:79-120).  All must match the torch/NumPy
 reference semantics EXACTLY (no tolerance)."""
 
-import numpy as np
 import pytest
 import torch
 
 pytestmark = pytest.mark.gpu
 
-from npairloss_amd.config.params import NPairLossConfig
 from npairloss_amd.ops import _backend
 from npairloss_amd.ops import npair_loss as NL
 
