@@ -22,7 +22,7 @@ def _worker(rank, world, tmpdir, cfg_idx, seed):
     os.environ.setdefault("MASTER_PORT", "29531")
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
-                from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
+        from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
 
         cfg = config_grid()[cfg_idx]
         f, lab = make_batch(num_classes=8, per_class=4, dim=32, seed=seed)
