@@ -50,8 +50,14 @@ def _worker(rank, world, tmpdir, cfg_idx, seed):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2, 4])
-@pytest.mark.parametrize("cfg_idx", [0, 3, len(config_grid()) - 1])
+def _world_grid():
+    # full config grid at world 2/4 + one 8-rank case mirroring the
+    # 8-GPU node topology (G = 8B; GLOBAL mining spans all ranks)
+    return ([(w, c) for w in (2, 4) for c in (0, 3, len(config_grid()) - 1)]
+            + [(8, 0)])
+
+
+@pytest.mark.parametrize("world,cfg_idx", _world_grid())
 def test_gloo_matches_multirank_oracle(world, cfg_idx):
     from npairloss_amd.ops import oracle
 
