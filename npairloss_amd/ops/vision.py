@@ -212,10 +212,14 @@ class _Conv1x1BiasReLUFn(torch.autograd.Function):
             N = w.shape[0]
             g4, db = C.biasrelu_bwd(y, dy)  # one pass: dReLU mask + bias grad
             gm = g4.permute(0, 2, 3, 1).reshape(B * H * W, N)
-            xm = x.permute(0, 2, 3, 1).reshape(B * H * W, K)
             wt = w.t().contiguous()  # tiny K x N copy
             dx = C.conv1x1_dgrad(gm, wt).view(B, H, W, K).permute(0, 3, 1, 2)
-            dw = (gm.t() @ xm).view(N, K, 1, 1)  # hipBLASLt TN, fp32 accum
+            # wgrad: MIOpen wgrad-only (split-K tuned; an unsplit library TN
+            # GEMM loses 10x on the M-site reduction — profiles/kb_conv1x1)
+            w4 = w.view(N, K, 1, 1)
+            dw = torch.ops.aten.convolution_backward(
+                g4, x, w4, None, (1, 1), (0, 0), (1, 1), False, (0, 0), 1,
+                [False, True, False])[1].view(N, K, 1, 1)
         else:
             mask = y > 0
             g = dy * mask
@@ -267,22 +271,27 @@ class _Conv1x1MM(torch.autograd.Function):
 
 
 class Conv1x1BiasReLU(nn.Module):
-    """1x1 conv + bias + ReLU.  Three GPU strategies (NPAIR_CONV1X1 env):
-      hybrid (default) — MIOpen forward GEMM + fused BiasReLU, backward =
-                         fused dReLU+bias-grad pass + two hipBLASLt GEMMs
-      custom           — our MFMA GEMM with bias+relu epilogue end to end
-                         (csrc/conv1x1.hip; currently slower than MIOpen)
-      off              — plain conv2d + fused BiasReLU (MIOpen both ways)
+    """1x1 conv + bias + ReLU.  GPU strategies (NPAIR_CONV1X1 env):
+      off (default) — plain conv2d + fused BiasReLU (MIOpen both ways;
+                      measured fastest overall this round)
+      hybrid        — MIOpen fwd + fused BiasReLU; backward = fused
+                      dReLU+bias-grad + our v2 MFMA dgrad + MIOpen wgrad
+      custom        — our MFMA GEMM with bias+relu epilogue forward, same
+                      backward chain (csrc/conv1x1.hip v2, 115-245 TF)
+      auto          — per-shape: custom where the fused v2 forward measured
+                      faster than MIOpen+BiasReLU (large M), hybrid below
+                      (profiles/kb_conv1x1_v2.log; round-3 default candidate)
     Exposes `.weight`/`.bias` for caffe_names() checkpoint IO."""
+
+    # fused v2 fwd wins at the 56^2 shape (M=802816) and ties ~28^2;
+    # MIOpen keeps the 14^2/7^2 tiles (kb_conv1x1_v2.log)
+    AUTO_MIN_SITES = 28 * 28 * 256
 
     def __init__(self, cin, cout):
         super().__init__()
         self.conv = nn.Conv2d(cin, cout, 1, bias=False)
         self.bias = nn.Parameter(torch.zeros(cout))
         import os
-        # default "off": measured fastest (profiles/kb_conv1x1 — MIOpen wins
-        # both directions at the bench shapes; hipBLASLt's unsplit TN wgrad
-        # loses badly on the M=800k reduction, our MFMA GEMM is mid-tuning)
         self.mode = os.environ.get("NPAIR_CONV1X1", "off")
 
     @property
@@ -290,8 +299,12 @@ class Conv1x1BiasReLU(nn.Module):
         return self.conv.weight
 
     def forward(self, x):
-        if self.mode == "custom":
+        mode = self.mode
+        if mode == "auto":
+            sites = x.shape[0] * x.shape[2] * x.shape[3] if x.dim() == 4 else 0
+            mode = "custom" if sites > self.AUTO_MIN_SITES else "hybrid"
+        if mode == "custom":
             return _Conv1x1BiasReLUFn.apply(x, self.conv.weight, self.bias)
-        if self.mode == "hybrid":
+        if mode == "hybrid":
             return _BiasReLUFn.apply(_Conv1x1MM.apply(x, self.conv.weight), self.bias)
         return _BiasReLUFn.apply(self.conv(x), self.bias)

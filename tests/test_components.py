@@ -507,7 +507,7 @@ def test_conv1x1_module_modes_cpu():
     from npairloss_amd.ops.vision import Conv1x1BiasReLU
 
     results = []
-    for mode in ("off", "hybrid", "custom"):
+    for mode in ("off", "hybrid", "custom", "auto"):
         torch.manual_seed(11)
         m = Conv1x1BiasReLU(8, 12)
         m.mode = mode
