@@ -189,3 +189,41 @@ def test_recall_strict_greater():
     lab_g = np.array([0, 0, 0])
     r = oracle.retrieval_recall(S, lab_l, lab_g, rank=0, top_k=1)
     assert r == 0.0
+
+
+def test_degenerate_batches_no_nan():
+    """Boundary shapes must produce finite guarded outputs, never NaN/crash:
+    B=1 (one query), all-same-label (empty negative lists), all-distinct
+    labels (empty positive lists)."""
+    import torch
+
+    from npairloss_amd.config.params import NPairLossConfig
+    from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
+
+    cfg = NPairLossConfig(margin_diff=-0.05, an_mining_region="LOCAL",
+                          an_mining_method="HARD", ap_mining_region="GLOBAL",
+                          ap_mining_method="RELATIVE_HARD")
+    mod = NPairMultiClassLoss(cfg)
+    torch.manual_seed(0)
+
+    # B=1: database of G-1=0 entries -> every term zero-guarded
+    f1 = torch.nn.functional.normalize(torch.randn(1, 16), dim=1).requires_grad_(True)
+    out = mod(f1, torch.tensor([0]))
+    assert torch.isfinite(out.loss)
+    out.loss.backward()
+    assert torch.isfinite(f1.grad).all()
+
+    # all same label: diff lists empty everywhere
+    f2 = torch.nn.functional.normalize(torch.randn(6, 16), dim=1).requires_grad_(True)
+    out = mod(f2, torch.zeros(6, dtype=torch.long))
+    assert torch.isfinite(out.loss)
+    out.loss.backward()
+    assert torch.isfinite(f2.grad).all()
+
+    # all distinct labels: positive lists empty -> loss 0 by the guards
+    f3 = torch.nn.functional.normalize(torch.randn(6, 16), dim=1).requires_grad_(True)
+    out = mod(f3, torch.arange(6))
+    assert torch.isfinite(out.loss)
+    assert float(out.loss) == 0.0  # no mined positives anywhere (.cu:162-169)
+    out.loss.backward()
+    assert torch.isfinite(f3.grad).all()
