@@ -520,3 +520,20 @@ def test_conv1x1_module_modes_cpu():
     for r in results[1:]:
         for a, b in zip(results[0], r):
             torch.testing.assert_close(a, b, rtol=1e-5, atol=1e-6)
+
+
+def test_device_synthetic_batches_structure():
+    """DeviceSyntheticBatches: P x K label structure, fixed shapes, cycles
+    with fresh permutations (bench-parity trainer input)."""
+    from npairloss_amd.data.synthetic import DeviceSyntheticBatches
+
+    src = DeviceSyntheticBatches(5, 2, image_size=32, num_classes=16,
+                                 device=torch.device("cpu"), n_distinct=3, seed=1)
+    seen = []
+    for x, lab in src:
+        assert x.shape == (10, 3, 32, 32)
+        counts = lab.bincount(minlength=16)
+        assert (counts[counts > 0] == 2).all()  # exactly K per identity
+        assert (counts > 0).sum() == 5          # P identities
+        seen.append(lab.clone())
+    assert len(seen) == 3
