@@ -61,3 +61,18 @@ def test_bench_torchrun_two_ranks():
     # exactly ONE json line (rank 0 only prints)
     njson = sum(1 for l in r.stdout.splitlines() if l.strip().startswith("{"))
     assert njson == 1
+
+
+def test_bench_data_pipeline_schema():
+    """--data-pipeline keeps the contract: same JSON schema, data field
+    marks the timed input pipeline."""
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "1", "--warmup", "0",
+         "--batch", "8", "--image", "64", "--data-pipeline"],
+        cwd=ROOT, capture_output=True, text=True, timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = _last_json_line(r.stdout)
+    for field in REQUIRED_FIELDS:
+        assert field in out, f"missing {field}"
+    assert out["data"] == "synthetic+pipeline"
+    assert out["config"]["data_pipeline"] is True
