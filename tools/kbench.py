@@ -108,6 +108,33 @@ def main():
               lambda dy=dy, idx=idx, s=stride, h=shape[2], w=shape[3]: C.maxpool3_bwd(dy, idx, s, h, w),
               bytes_moved=nb + y.numel() * 3)
 
+    # fused conv-unit kernels (round 2)
+    for name, (Cc, H) in (("56^2x64", (64, 56)), ("28^2x256", (256, 28)),
+                          ("7^2x832", (832, 7))):
+        xx = cl(torch.randn(256, Cc, H, H, device=dev, dtype=torch.bfloat16))
+        bb = torch.randn(Cc, device=dev)
+        nb = xx.numel() * 2
+        yy = C.biasrelu_fwd(xx, bb)
+        dyy = torch.randn_like(yy)
+        bench(f"biasrelu_fwd {name}", lambda xx=xx, bb=bb: C.biasrelu_fwd(xx, bb),
+              bytes_moved=2 * nb)
+        bench(f"biasrelu_bwd {name}", lambda yy=yy, dyy=dyy: C.biasrelu_bwd(yy, dyy),
+              bytes_moved=3 * nb)
+
+    for name, (K, H, N) in (("conv1x1 56^2 K64 N64", (64, 56, 64)),
+                            ("conv1x1 28^2 K192 N128", (192, 28, 128)),
+                            ("conv1x1 7^2 K832 N384", (832, 7, 384))):
+        M = 256 * H * H
+        xm = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        ww = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.05
+        bb = torch.randn(N, device=dev)
+        fl = 2.0 * M * N * K
+        bench(f"{name} fused v2 fwd", lambda xm=xm, ww=ww, bb=bb:
+              C.conv1x1_bias_relu_fwd(xm, ww, bb), flops=fl)
+        gg = torch.randn(M, N, device=dev, dtype=torch.bfloat16)
+        wt = ww.t().contiguous()
+        bench(f"{name} dgrad v2", lambda gg=gg, wt=wt: C.conv1x1_dgrad(gg, wt), flops=fl)
+
     for name, shape in (("norm1 112^2x64", (256, 64, 112, 112)),
                         ("norm2 56^2x192", (256, 192, 56, 56))):
         xx = cl(torch.randn(*shape, device=dev, dtype=torch.bfloat16))
