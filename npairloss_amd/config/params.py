@@ -130,6 +130,7 @@ class SolverConfig:
     gamma: float = 1.0
     power: float = 1.0
     max_iter: int = 0
+    iter_size: int = 1
     momentum: float = 0.0
     weight_decay: float = 0.0
     snapshot: int = 0

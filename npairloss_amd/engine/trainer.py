@@ -196,11 +196,10 @@ class Trainer:
             images = images.to(torch.bfloat16)
         return images
 
-    def _compute_step(self, images: torch.Tensor, labels: torch.Tensor):
-        """Device-only step body (no host reads): forward + loss + backward
-        + grad comm + optimizer.  hipGraph-capturable — inputs must already
-        be prepared (_prepare) and sit at fixed addresses during capture."""
-        self.reducer.zero_grad()  # zeroes the persistent flat grad buckets
+    def _micro_step(self, images: torch.Tensor, labels: torch.Tensor,
+                    loss_scale: float = 1.0):
+        """Forward + loss + (scaled) backward only — gradients accumulate
+        into the flat buckets; no zeroing, comm wait, or optimizer here."""
         if self.amp_dtype is not None and self.device.type == "cuda":
             with torch.autocast("cuda", dtype=self.amp_dtype):
                 feats = self.model(images)
@@ -208,21 +207,57 @@ class Trainer:
         else:
             feats = self.model(images)
             out = self.loss(feats, labels)
-        out.loss.backward()
+        loss = out.loss if loss_scale == 1.0 else out.loss * loss_scale
+        loss.backward()
+        return out
+
+    def _compute_step(self, images: torch.Tensor, labels: torch.Tensor):
+        """Device-only step body (no host reads): forward + loss + backward
+        + grad comm + optimizer.  hipGraph-capturable — inputs must already
+        be prepared (_prepare) and sit at fixed addresses during capture."""
+        self.reducer.zero_grad()  # zeroes the persistent flat grad buckets
+        out = self._micro_step(images, labels)
         self.reducer.finalize()
         self.optimizer.step()
         return out
 
-    def train_step(self, images: torch.Tensor, labels: torch.Tensor) -> dict:
+    def train_step(self, images: torch.Tensor, labels: torch.Tensor,
+                   extra_batches: Optional[list] = None) -> dict:
+        """One solver iteration.  With solver.iter_size > 1 (Caffe's
+        gradient accumulation), `extra_batches` carries the remaining
+        iter_size-1 micro-batches: losses are scaled by 1/iter_size and
+        gradients accumulate in the flat buckets (no communication) until
+        the final micro-batch, exactly like Caffe's normalized update."""
         self.model.train()
         images = images.to(self.device, non_blocking=True)
         labels = labels.to(self.device, non_blocking=True)
         lr = self.solver.lr_at(self.iter)
         self.optimizer.set_lr(lr)
+        micro = [(images, labels)] + list(extra_batches or [])
+        n_micro = max(len(micro), 1)
         with self.timers.phase("data_prep"):
             images = self._prepare(images)
         with self.timers.phase("step"):
-            out = self._compute_step(images, labels)
+            if n_micro == 1:
+                out = self._compute_step(images, labels)
+            else:
+                self.reducer.zero_grad()
+                self.reducer.set_accumulate(True)
+                try:
+                    for i, (mx, mlab) in enumerate(micro):
+                        if i > 0:
+                            mx = self._prepare(mx.to(self.device, non_blocking=True))
+                            mlab = mlab.to(self.device, non_blocking=True)
+                        else:
+                            mx = images
+                            mlab = labels
+                        if i == n_micro - 1:
+                            self.reducer.set_accumulate(False)
+                        out = self._micro_step(mx, mlab, 1.0 / n_micro)
+                finally:
+                    self.reducer.set_accumulate(False)
+                self.reducer.finalize()
+                self.optimizer.step()
         self.iter += 1
         if self.divergence_check and self.iter % self.divergence_check == 0:
             lv = float(out.loss.detach())
@@ -358,16 +393,27 @@ class Trainer:
         epoch = 0
         data_iter = iter(self.train_loader)
         while self.iter < max_iter:
-            try:
-                images, labels = next(data_iter)
-            except StopIteration:
-                epoch += 1
-                bs = getattr(self.train_loader, "batch_sampler", None)
-                if bs is not None and hasattr(bs, "set_epoch"):
-                    bs.set_epoch(epoch)  # deterministic (base_seed, epoch) reseed
-                data_iter = iter(self.train_loader)
-                images, labels = next(data_iter)
-            if self.use_graph and self.device.type == "cuda":
+            def next_batch():
+                nonlocal data_iter, epoch
+                try:
+                    return next(data_iter)
+                except StopIteration:
+                    epoch += 1
+                    bs = getattr(self.train_loader, "batch_sampler", None)
+                    if bs is not None and hasattr(bs, "set_epoch"):
+                        bs.set_epoch(epoch)  # deterministic (base_seed, epoch)
+                    data_iter = iter(self.train_loader)
+                    return next(data_iter)
+
+            images, labels = next_batch()
+            iter_size = max(1, self.solver.iter_size)
+            if iter_size > 1:
+                # Caffe gradient accumulation: iter_size micro-batches per
+                # solver iteration (always eager — the captured graph holds
+                # exactly one accumulate+step sequence)
+                extra = [next_batch() for _ in range(iter_size - 1)]
+                stats = self.train_step(images, labels, extra_batches=extra)
+            elif self.use_graph and self.device.type == "cuda":
                 stats = self._graph_step(images, labels)
             else:
                 stats = self.train_step(images, labels)

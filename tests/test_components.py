@@ -537,3 +537,42 @@ def test_device_synthetic_batches_structure():
         assert (counts > 0).sum() == 5          # P identities
         seen.append(lab.clone())
     assert len(seen) == 3
+
+
+def test_iter_size_gradient_accumulation_matches_caffe_semantics():
+    """solver.iter_size = 2 (Caffe gradient accumulation): one optimizer
+    step whose gradient equals the MEAN over the micro-batch gradients
+    (Caffe normalizes by iter_size), communicated once."""
+    import torch.nn as nn
+
+    from npairloss_amd.config.params import NPairLossConfig, SolverConfig
+    from npairloss_amd.engine.trainer import Trainer
+    from npairloss_amd.ops.npair_loss import NPairMultiClassLoss
+
+    torch.manual_seed(0)
+    ba = (torch.randn(4, 8), torch.tensor([0, 0, 1, 1]))
+    bb = (torch.randn(4, 8), torch.tensor([2, 2, 3, 3]))
+
+    def make():
+        torch.manual_seed(1)
+        return nn.Sequential(nn.Linear(8, 16), nn.ReLU(), nn.Linear(16, 8))
+
+    # accumulated run: iter_size 2 via the trainer
+    solver = SolverConfig(base_lr=0.1, momentum=0.9, iter_size=2, max_iter=1)
+    tr = Trainer(make(), NPairMultiClassLoss(NPairLossConfig()), solver,
+                 train_loader=[ba, bb], device=torch.device("cpu"))
+    tr.fit(max_iter=1)
+    assert tr.iter == 1  # two micro-batches = ONE solver iteration
+
+    # manual reference: mean gradient of the two micro-batches, one step
+    ref = make()
+    loss_mod = NPairMultiClassLoss(NPairLossConfig())
+    for p in ref.parameters():
+        p.grad = torch.zeros_like(p)
+    for x, lab in (ba, bb):
+        (loss_mod(ref(x), lab).loss / 2).backward()
+    from npairloss_amd.engine.solver import CaffeSGD
+    opt = CaffeSGD(ref.parameters(), lr=0.1, momentum=0.9)
+    opt.step()
+    for pa, pb in zip(tr.model.parameters(), ref.parameters()):
+        torch.testing.assert_close(pa, pb, rtol=1e-6, atol=1e-7)
